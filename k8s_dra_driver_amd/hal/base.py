@@ -1,0 +1,107 @@
+"""Hardware abstraction interface (``deviceLib`` analog, ``nvlib.go:44-109``).
+
+Two implementations:
+
+- :class:`k8s_dra_driver_amd.hal.amdsmi.AmdSmiDeviceLib` — libamd_smi via the
+  in-tree C++ extension plus KFD/DRM sysfs (the NVML-replacement native
+  boundary),
+- :class:`k8s_dra_driver_amd.hal.fake.FakeDeviceLib` — an in-memory
+  8xMI355X node with a full partition state machine and fault injection
+  (the test backend the reference never had).
+
+Unlike the reference — which re-runs NVML Init/Shutdown around every call
+(``nvlib.go:96-109``) — a DeviceLib is opened once and kept; enumeration can
+be refreshed cheaply, which the dynamic-repartition path depends on.
+"""
+
+from __future__ import annotations
+
+import abc
+from typing import Dict, List, Optional
+
+from .model import GpuInfo
+
+
+class HalError(RuntimeError):
+    """Base error for hardware-layer failures."""
+
+
+class HalUnavailable(HalError):
+    """The native backend cannot run here (no libamd_smi / no KFD)."""
+
+
+class HalNotSupported(HalError):
+    """Operation not supported by this device/backend."""
+
+
+class DeviceLib(abc.ABC):
+    """Synchronous hardware operations on one node's GPUs."""
+
+    # -- lifecycle ---------------------------------------------------------
+    @abc.abstractmethod
+    def open(self) -> None:
+        """Bind the backend (amdsmi_init analog). Idempotent."""
+
+    @abc.abstractmethod
+    def close(self) -> None:
+        """Release the backend (amdsmi_shut_down analog). Idempotent."""
+
+    def __enter__(self) -> "DeviceLib":
+        self.open()
+        return self
+
+    def __exit__(self, *exc) -> None:
+        self.close()
+
+    # -- enumeration -------------------------------------------------------
+    @abc.abstractmethod
+    def enumerate(self) -> List[GpuInfo]:
+        """Snapshot every physical GPU incl. current partitions & topology.
+
+        The ``enumerateAllPossibleDevices`` analog (``nvlib.go:111-136``) —
+        but re-runnable: callers refresh after repartition instead of
+        restarting the plugin (a reference limitation).
+        """
+
+    # -- partitioning (dynamic MIG analog; real on MI355X) ------------------
+    @abc.abstractmethod
+    def set_compute_partition(self, gpu_index: int, mode: str) -> None:
+        """Switch SPX/DPX/QPX/CPX (``amdsmi_set_gpu_compute_partition``).
+
+        Affects the whole GPU; caller (partition.manager) must hold the GPU
+        ownership lock and have drained claims first.
+        """
+
+    @abc.abstractmethod
+    def set_memory_partition(self, gpu_index: int, mode: str) -> None:
+        """Switch NPS mode (``amdsmi_set_gpu_memory_partition``).
+
+        May require no processes on the GPU; can be slow (driver reload
+        semantics on some stacks).
+        """
+
+    # -- scheduler / sharing controls (N7/N8 analogs) ------------------------
+    @abc.abstractmethod
+    def set_timeslice_quantum(self, gpu_index: int, quantum_us: Optional[int]) -> None:
+        """Best-effort per-GPU scheduler quantum (the ``nvidia-smi
+        compute-policy --set-timeslice`` analog, ``nvlib.go:521-539``).
+
+        On amdgpu the queue-scheduling quantum is a module-level control;
+        backends record the request and apply what the platform allows.
+        ``None`` restores the default.
+        """
+
+    # -- device nodes for CDI ------------------------------------------------
+    @abc.abstractmethod
+    def device_node_paths(self, gpu_index: int, partition_id: Optional[int] = None) -> Dict[str, str]:
+        """Paths of the device nodes a claim must inject.
+
+        Returns a dict with keys ``kfd``, ``renderD``, ``card`` mapped to
+        absolute /dev paths (the /dev/nvidia* analog, reference
+        ``cdi.go:158-227`` via nvcdi).
+        """
+
+    # -- health --------------------------------------------------------------
+    @abc.abstractmethod
+    def health_check(self, gpu_index: int) -> Dict[str, str]:
+        """Lightweight liveness/metrics probe for failure detection."""

@@ -1,0 +1,294 @@
+"""Device information model and ResourceSlice attribute projection.
+
+The analog of the reference's info structs + attribute conversion
+(``cmd/nvidia-dra-plugin/deviceinfo.go:30-223``): every allocatable device —
+a whole MI355X in SPX, or one compute partition in DPX/QPX/CPX — projects to
+a ``resource.k8s.io/v1beta1`` Device entry with typed attributes that CEL
+selectors can match and capacities the scheduler can count.
+
+Naming (identity survives repartition — SURVEY.md §7 hard-part 3):
+
+- whole GPU:            ``gpu-<index>``             (reference: ``gpu-0``)
+- compute partition:    ``gpu-<index>-<mode>-<pid>`` (e.g. ``gpu-0-cpx-3``;
+  reference MIG analog: ``gpu-0-mig-9-4-4``, ``deviceinfo.go:87-96``)
+
+The name is keyed on (parent GPU index, partition mode, partition slot) — all
+stable across repartition cycles — never on mutable DRM minors.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from ..partition.catalog import (
+    GFX950_ARCH,
+    GFX950_XGMI_LINKS_PER_GPU,
+    PartitionProfile,
+)
+
+#: attribute/capacity qualified-name domain (reference uses gpu.nvidia.com)
+DOMAIN = "gpu.amd.com"
+
+
+def qualified(name: str) -> str:
+    return f"{DOMAIN}/{name}"
+
+
+def quantity_mib(mib: int) -> str:
+    """Render a Mi quantity the way kubelet/scheduler expect (BinarySI)."""
+    if mib % (1024 * 1024) == 0:
+        return f"{mib // (1024 * 1024)}Ti"
+    if mib % 1024 == 0:
+        return f"{mib // 1024}Gi"
+    return f"{mib}Mi"
+
+
+@dataclass(frozen=True)
+class XgmiLink:
+    """One point-to-point xGMI link to a peer GPU.
+
+    MI355X xGMI is point-to-point (no switch): 7 links x ~153 GB/s per GPU on
+    an 8-GPU node. Published per-device so CEL selectors and the allocator can
+    score claim placements by mutual adjacency.
+    """
+
+    peer_oam_id: int
+    peer_uuid: str
+    num_lanes: int = 16
+    max_bandwidth_gbps: int = 153
+
+
+@dataclass
+class PartitionedDeviceInfo:
+    """One compute partition of a GPU (allocatable device in DPX/QPX/CPX).
+
+    MIG-device analog (``deviceinfo.go:44-63``): each partition is its own
+    KFD node with its own DRM render node while the mode is active.
+    """
+
+    parent_index: int
+    parent_uuid: str
+    partition_id: int
+    profile: PartitionProfile
+    kfd_node_id: int = -1
+    render_minor: int = -1
+    card_minor: int = -1
+
+    @property
+    def canonical_name(self) -> str:
+        return (
+            f"gpu-{self.parent_index}-"
+            f"{self.profile.compute_mode.lower()}-{self.partition_id}"
+        )
+
+    @property
+    def uuid(self) -> str:
+        """Stable partition identity: parent UUID x mode x slot."""
+        return (
+            f"{self.parent_uuid}-"
+            f"{self.profile.compute_mode.lower()}-{self.partition_id}"
+        )
+
+
+@dataclass
+class GpuInfo:
+    """One physical MI355X (``deviceinfo.go:30-42`` analog, MI355X attrs)."""
+
+    index: int
+    uuid: str
+    oam_id: int
+    product_name: str = "AMD Instinct MI355X"
+    architecture: str = GFX950_ARCH
+    device_id: int = 0x75A0
+    pcie_bdf: str = ""
+    vram_total_mib: int = 288 * 1024
+    vram_type: str = "HBM3E"
+    cu_count: int = 256
+    xcd_count: int = 8
+    simd_count: int = 1024
+    driver_version: str = ""
+    rocm_version: str = ""
+    kfd_node_id: int = -1
+    render_minor: int = -1
+    card_minor: int = -1
+    compute_partition: str = "SPX"
+    memory_partition: str = "NPS1"
+    nps_caps: List[str] = field(default_factory=lambda: ["NPS1", "NPS4"])
+    compute_caps: List[str] = field(
+        default_factory=lambda: ["SPX", "DPX", "QPX", "CPX"]
+    )
+    xgmi_hive_id: str = ""
+    xgmi_node_id: int = -1
+    links: List[XgmiLink] = field(default_factory=list)
+    #: partitions present when compute_partition != SPX
+    partitions: List[PartitionedDeviceInfo] = field(default_factory=list)
+
+    @property
+    def canonical_name(self) -> str:
+        return f"gpu-{self.index}"
+
+    def xgmi_peer_oam_ids(self) -> List[int]:
+        return sorted(l.peer_oam_id for l in self.links)
+
+
+# ---------------------------------------------------------------------------
+# ResourceSlice Device projection (resource.k8s.io/v1beta1 JSON shape)
+# ---------------------------------------------------------------------------
+
+
+def _attr_str(v: str) -> Dict[str, str]:
+    return {"string": v}
+
+
+def _attr_int(v: int) -> Dict[str, int]:
+    return {"int": v}
+
+
+def _attr_bool(v: bool) -> Dict[str, bool]:
+    return {"bool": v}
+
+
+def _attr_ver(v: str) -> Dict[str, str]:
+    return {"version": v}
+
+
+def _common_gpu_attrs(gpu: GpuInfo) -> Dict[str, dict]:
+    """Attributes shared by the whole GPU and its partitions."""
+    attrs = {
+        qualified("productName"): _attr_str(gpu.product_name),
+        qualified("architecture"): _attr_str(gpu.architecture),
+        qualified("brand"): _attr_str("Instinct"),
+        qualified("oamId"): _attr_int(gpu.oam_id),
+        qualified("pcieBDF"): _attr_str(gpu.pcie_bdf),
+        qualified("vramType"): _attr_str(gpu.vram_type),
+        qualified("xgmiHiveId"): _attr_str(gpu.xgmi_hive_id),
+        qualified("xgmiNodeId"): _attr_int(gpu.xgmi_node_id),
+        qualified("xgmiLinkCount"): _attr_int(len(gpu.links)),
+        qualified("xgmiPeerOamIds"): _attr_str(
+            ",".join(str(i) for i in gpu.xgmi_peer_oam_ids())
+        ),
+    }
+    if gpu.driver_version:
+        attrs[qualified("driverVersion")] = _attr_ver(gpu.driver_version)
+    if gpu.rocm_version:
+        attrs[qualified("rocmVersion")] = _attr_ver(gpu.rocm_version)
+    return attrs
+
+
+def gpu_to_device(gpu: GpuInfo) -> dict:
+    """Project a whole GPU (SPX) to a v1beta1 Device (``deviceinfo.go:98-142``)."""
+    attrs = _common_gpu_attrs(gpu)
+    attrs.update(
+        {
+            qualified("type"): _attr_str("gpu"),
+            qualified("uuid"): _attr_str(gpu.uuid),
+            qualified("index"): _attr_int(gpu.index),
+            qualified("kfdNodeId"): _attr_int(gpu.kfd_node_id),
+            qualified("renderDMinor"): _attr_int(gpu.render_minor),
+            qualified("computePartition"): _attr_str(gpu.compute_partition),
+            qualified("memoryPartition"): _attr_str(gpu.memory_partition),
+            qualified("partitionable"): _attr_bool(len(gpu.compute_caps) > 1),
+        }
+    )
+    capacity = {
+        qualified("memory"): {"value": quantity_mib(gpu.vram_total_mib)},
+        qualified("computeUnits"): {"value": str(gpu.cu_count)},
+        qualified("xcds"): {"value": str(gpu.xcd_count)},
+    }
+    return {
+        "name": gpu.canonical_name,
+        "basic": {"attributes": attrs, "capacity": capacity},
+    }
+
+
+def partition_to_device(gpu: GpuInfo, part: PartitionedDeviceInfo) -> dict:
+    """Project one compute partition (``deviceinfo.go:144-206`` MIG analog).
+
+    Publishes ``parentUUID`` so claims can constrain partitions to one die
+    via matchAttribute (reference demo ``gpu-test4.yaml:42-44``), plus the
+    memory-domain id and memorySliceN capacities so co-domain placement is
+    expressible.
+    """
+    prof = part.profile
+    attrs = _common_gpu_attrs(gpu)
+    attrs.update(
+        {
+            qualified("type"): _attr_str("partition"),
+            qualified("uuid"): _attr_str(part.uuid),
+            qualified("parentUUID"): _attr_str(gpu.uuid),
+            qualified("parentIndex"): _attr_int(gpu.index),
+            qualified("index"): _attr_int(part.partition_id),
+            qualified("partitionId"): _attr_int(part.partition_id),
+            qualified("computePartition"): _attr_str(prof.compute_mode),
+            qualified("memoryPartition"): _attr_str(prof.memory_mode),
+            qualified("memoryDomain"): _attr_int(
+                prof.memory_domain_of(part.partition_id)
+            ),
+            qualified("kfdNodeId"): _attr_int(part.kfd_node_id),
+            qualified("renderDMinor"): _attr_int(part.render_minor),
+        }
+    )
+    capacity = {
+        qualified("memory"): {
+            "value": quantity_mib(prof.memory_mib_per_partition)
+        },
+        qualified("computeUnits"): {"value": str(prof.cus_per_partition)},
+        qualified("xcds"): {"value": str(prof.xcds_per_partition)},
+    }
+    for s in prof.memory_slices_of(part.partition_id):
+        capacity[qualified(f"memorySlice{s}")] = {"value": "1"}
+    return {
+        "name": part.canonical_name,
+        "basic": {"attributes": attrs, "capacity": capacity},
+    }
+
+
+@dataclass
+class AllocatableDevice:
+    """Tagged union of allocatable device kinds (``allocatable.go:29-108``)."""
+
+    gpu: Optional[GpuInfo] = None
+    partition: Optional[PartitionedDeviceInfo] = None
+    _parent: Optional[GpuInfo] = None
+
+    @classmethod
+    def from_gpu(cls, gpu: GpuInfo) -> "AllocatableDevice":
+        return cls(gpu=gpu)
+
+    @classmethod
+    def from_partition(
+        cls, gpu: GpuInfo, part: PartitionedDeviceInfo
+    ) -> "AllocatableDevice":
+        return cls(partition=part, _parent=gpu)
+
+    @property
+    def kind(self) -> str:
+        return "gpu" if self.gpu is not None else "partition"
+
+    @property
+    def canonical_name(self) -> str:
+        if self.gpu is not None:
+            return self.gpu.canonical_name
+        assert self.partition is not None
+        return self.partition.canonical_name
+
+    @property
+    def uuid(self) -> str:
+        if self.gpu is not None:
+            return self.gpu.uuid
+        assert self.partition is not None
+        return self.partition.uuid
+
+    @property
+    def parent_gpu(self) -> GpuInfo:
+        if self.gpu is not None:
+            return self.gpu
+        assert self._parent is not None
+        return self._parent
+
+    def to_device(self) -> dict:
+        if self.gpu is not None:
+            return gpu_to_device(self.gpu)
+        assert self.partition is not None and self._parent is not None
+        return partition_to_device(self._parent, self.partition)

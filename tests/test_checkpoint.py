@@ -1,0 +1,79 @@
+"""Checkpoint tests: per-claim files, checksums, recovery scan."""
+
+import json
+import os
+
+import pytest
+
+from k8s_dra_driver_amd.state.checkpoint import (
+    CheckpointCorrupt,
+    CheckpointStore,
+    PreparedClaim,
+    PreparedDevice,
+)
+
+
+def _claim(uid="uid-1"):
+    return PreparedClaim(
+        claim_uid=uid,
+        namespace="default",
+        name="claim-a",
+        devices=[
+            PreparedDevice(
+                request_names=["gpu"],
+                pool_name="node-1",
+                device_name="gpu-0",
+                cdi_device_ids=["k8s.gpu.amd.com/device=gpu-0"],
+                parent_gpu_index=0,
+                kind="gpu",
+                device_uuid="amd-mi355x-00",
+            )
+        ],
+        sharing_strategy="TimeSlicing",
+        timeslice_gpus=[0],
+    )
+
+
+def test_roundtrip(tmp_path):
+    store = CheckpointStore(str(tmp_path))
+    store.write(_claim())
+    got = store.read("uid-1")
+    assert got.claim_uid == "uid-1"
+    assert got.devices[0].device_name == "gpu-0"
+    assert got.timeslice_gpus == [0]
+
+
+def test_absent_returns_none(tmp_path):
+    assert CheckpointStore(str(tmp_path)).read("nope") is None
+
+
+def test_checksum_detects_tamper(tmp_path):
+    store = CheckpointStore(str(tmp_path))
+    store.write(_claim())
+    path = os.path.join(str(tmp_path), "claims", "uid-1.json")
+    obj = json.load(open(path))
+    obj["v1"]["devices"][0]["device_name"] = "gpu-7"
+    json.dump(obj, open(path, "w"))
+    with pytest.raises(CheckpointCorrupt, match="checksum"):
+        store.read("uid-1")
+
+
+def test_per_claim_files_are_independent(tmp_path):
+    store = CheckpointStore(str(tmp_path))
+    store.write(_claim("a"))
+    store.write(_claim("b"))
+    files = os.listdir(tmp_path / "claims")
+    assert sorted(files) == ["a.json", "b.json"]
+    store.delete("a")
+    assert store.read("a") is None
+    assert store.read("b") is not None
+
+
+def test_recovery_scan(tmp_path):
+    store = CheckpointStore(str(tmp_path))
+    for uid in ("a", "b", "c"):
+        store.write(_claim(uid))
+    fresh = CheckpointStore(str(tmp_path))
+    recovered = fresh.list_all()
+    assert sorted(recovered) == ["a", "b", "c"]
+    assert recovered["b"].devices[0].pool_name == "node-1"

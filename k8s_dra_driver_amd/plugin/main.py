@@ -1,0 +1,135 @@
+"""Kubelet plugin CLI entry point (reference cmd/nvidia-dra-plugin/main.go).
+
+Every flag mirrors an env var (reference main.go:73-123); Helm wires them
+from chart values. Runs until SIGTERM/SIGINT, then unpublishes slices and
+removes sockets.
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import signal
+import threading
+
+from .. import DRIVER_NAME
+from ..hal import new_device_lib
+from ..kube.client import InMemoryKube
+from ..metrics.prom import PluginMetrics
+from .driver import Driver
+from .server import PluginServer
+
+log = logging.getLogger(__name__)
+
+DEFAULT_PLUGIN_ROOT = "/var/lib/kubelet/plugins"
+DEFAULT_REGISTRY = "/var/lib/kubelet/plugins_registry"
+DEFAULT_CDI_ROOT = "/var/run/cdi"
+
+
+def _env(name: str, default: str = "") -> str:
+    return os.environ.get(name, default)
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser("amd-dra-kubeletplugin")
+    p.add_argument(
+        "--node-name",
+        default=_env("NODE_NAME", os.uname().nodename),
+        help="name of the node this plugin runs on",
+    )
+    p.add_argument(
+        "--hal",
+        default=_env("DRA_HAL", "amdsmi"),
+        choices=["amdsmi", "fake"],
+        help="hardware backend (fake = 8xMI355X model for dev clusters)",
+    )
+    p.add_argument(
+        "--cdi-root", default=_env("CDI_ROOT", DEFAULT_CDI_ROOT)
+    )
+    p.add_argument(
+        "--plugin-registration-path",
+        default=_env("PLUGIN_REGISTRATION_PATH", DEFAULT_REGISTRY),
+    )
+    p.add_argument(
+        "--plugin-path",
+        default=_env(
+            "PLUGIN_PATH", os.path.join(DEFAULT_PLUGIN_ROOT, DRIVER_NAME)
+        ),
+    )
+    p.add_argument(
+        "--kubeconfig",
+        default=_env("KUBECONFIG", ""),
+        help="kubeconfig path ('' = in-cluster; 'memory' = in-memory fake)",
+    )
+    p.add_argument(
+        "--kube-api-qps", type=float, default=float(_env("KUBE_API_QPS", "50"))
+    )
+    p.add_argument(
+        "--kube-api-burst", type=int, default=int(_env("KUBE_API_BURST", "100"))
+    )
+    p.add_argument(
+        "--metrics-port",
+        type=int,
+        default=int(_env("METRICS_PORT", "0")),
+        help="serve Prometheus /metrics on this port (0 = disabled)",
+    )
+    p.add_argument("-v", "--verbosity", type=int, default=int(_env("LOG_LEVEL", "1")))
+    return p
+
+
+def make_kube_client(args):
+    if args.kubeconfig == "memory":
+        return InMemoryKube()
+    from ..kube.http_kube import HttpKube
+
+    return HttpKube(
+        kubeconfig=args.kubeconfig or None,
+        qps=args.kube_api_qps,
+        burst=args.kube_api_burst,
+    )
+
+
+def main(argv=None) -> int:
+    args = build_parser().parse_args(argv)
+    logging.basicConfig(
+        level=logging.DEBUG if args.verbosity >= 4 else logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s %(message)s",
+    )
+    lib = new_device_lib(args.hal)
+    lib.open()
+    kube = make_kube_client(args)
+    metrics = PluginMetrics()
+    if args.metrics_port:
+        metrics.serve(args.metrics_port)
+    driver = Driver(
+        lib,
+        kube,
+        node_name=args.node_name,
+        cdi_root=args.cdi_root,
+        checkpoint_root=os.path.join(args.plugin_path, "state"),
+        metrics=metrics,
+    )
+    driver.startup()
+    server = PluginServer(
+        driver,
+        plugin_dir=args.plugin_path,
+        registry_dir=args.plugin_registration_path,
+    )
+    server.start()
+
+    stop = threading.Event()
+    for sig in (signal.SIGTERM, signal.SIGINT):
+        signal.signal(sig, lambda *_: stop.set())
+    log.info(
+        "amd-dra-kubeletplugin ready: node=%s hal=%s", args.node_name, args.hal
+    )
+    stop.wait()
+    server.stop()
+    driver.shutdown()
+    lib.close()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -93,6 +93,10 @@ class DeviceState:
             lib, in_use_fn=self.claims_holding_gpu
         )
 
+        #: optional callback fired after the allocatable set changes
+        #: (repartition) — the Driver republishes ResourceSlices from it.
+        self.on_allocatable_change = None
+
         self._registry_lock = threading.Lock()
         self._claim_locks: Dict[str, threading.Lock] = {}
         #: gpu_index -> set of claim uids with prepared devices on it
@@ -119,6 +123,10 @@ class DeviceState:
                 devices[d.canonical_name] = d
         with self._registry_lock:
             self._allocatable = devices
+        # Publisher fingerprints the device set, so an unchanged snapshot
+        # costs no API calls.
+        if self.on_allocatable_change is not None:
+            self.on_allocatable_change()
 
     def allocatable_devices(self) -> List[AllocatableDevice]:
         with self._registry_lock:

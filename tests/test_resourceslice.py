@@ -1,0 +1,73 @@
+"""ResourceSlice publisher reconcile tests."""
+
+from k8s_dra_driver_amd.kube.client import InMemoryKube
+from k8s_dra_driver_amd.kube.resourceslice import (
+    MAX_DEVICES_PER_SLICE,
+    ResourceSlicePublisher,
+)
+
+
+def _dev(name):
+    return {"name": name, "basic": {"attributes": {}, "capacity": {}}}
+
+
+def _pub(kube):
+    return ResourceSlicePublisher(
+        kube, driver_name="gpu.amd.com", node_name="node-a"
+    )
+
+
+def test_publish_creates_slice():
+    kube = InMemoryKube()
+    pub = _pub(kube)
+    pub.publish([_dev("gpu-0"), _dev("gpu-1")])
+    slices = kube.list_resource_slices("gpu.amd.com")
+    assert len(slices) == 1
+    assert slices[0]["spec"]["pool"]["generation"] == 1
+    assert len(slices[0]["spec"]["devices"]) == 2
+
+
+def test_unchanged_publish_is_noop():
+    kube = InMemoryKube()
+    pub = _pub(kube)
+    devs = [_dev("gpu-0")]
+    pub.publish(devs)
+    rv1 = kube.list_resource_slices("gpu.amd.com")[0]["metadata"]["resourceVersion"]
+    pub.publish(devs)
+    rv2 = kube.list_resource_slices("gpu.amd.com")[0]["metadata"]["resourceVersion"]
+    assert rv1 == rv2
+
+
+def test_changed_devices_bump_generation():
+    kube = InMemoryKube()
+    pub = _pub(kube)
+    pub.publish([_dev("gpu-0")])
+    pub.publish([_dev("gpu-0-cpx-0"), _dev("gpu-0-cpx-1")])
+    s = kube.list_resource_slices("gpu.amd.com")[0]
+    assert s["spec"]["pool"]["generation"] == 2
+    assert len(s["spec"]["devices"]) == 2
+
+
+def test_chunking_over_max_devices():
+    kube = InMemoryKube()
+    pub = _pub(kube)
+    many = [_dev(f"d-{i}") for i in range(MAX_DEVICES_PER_SLICE + 5)]
+    pub.publish(many)
+    slices = kube.list_resource_slices("gpu.amd.com")
+    assert len(slices) == 2
+    assert all(
+        s["spec"]["pool"]["resourceSliceCount"] == 2 for s in slices
+    )
+    total = sum(len(s["spec"]["devices"]) for s in slices)
+    assert total == len(many)
+    # shrink back to one slice: surplus slice must be deleted
+    pub.publish(many[:3])
+    assert len(kube.list_resource_slices("gpu.amd.com")) == 1
+
+
+def test_unpublish_all():
+    kube = InMemoryKube()
+    pub = _pub(kube)
+    pub.publish([_dev("gpu-0")])
+    pub.unpublish_all()
+    assert kube.list_resource_slices("gpu.amd.com") == []

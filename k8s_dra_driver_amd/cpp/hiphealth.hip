@@ -1,0 +1,244 @@
+// _hiphealth — gfx950 GPU health-check & demo workload kernels.
+//
+// The reference ships NVIDIA's prebuilt nbody sample as its demo/sharing
+// workload (demo/specs/quickstart/gpu-test5.yaml:53-87); this module is the
+// MI355X-native equivalent: a small set of CDNA4 kernels
+// the driver's GPU tests, smoke path and sharing demos run on prepared
+// devices:
+//
+//   bandwidth_gbs()  - float4 streaming copy (HBM3E bandwidth probe)
+//   mfma_check()     - v_mfma_f32_16x16x32_bf16 correctness (matrix pipes)
+//   mfma_tflops()    - bf16 MFMA throughput burn (8 independent
+//                      accumulators to cover the ~17 cyc/SIMD issue rate)
+//   burn_ms()        - VALU occupancy burner for time-slice/share demos
+//
+// Written for gfx950: 64-wide waves, 256 CUs in 8 XCDs (grids sized
+// >> 256 workgroups), 16 B/lane streaming accesses.
+// Build: hipcc --offload-arch=gfx950 (driven by setup.py / build()).
+
+#include <hip/hip_runtime.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                    \
+  do {                                                                     \
+    hipError_t _e = (expr);                                                \
+    if (_e != hipSuccess)                                                  \
+      throw std::runtime_error(std::string(#expr) + ": " +                 \
+                               hipGetErrorString(_e));                     \
+  } while (0)
+
+namespace {
+
+// ---------------------------------------------------------------------------
+// bandwidth: float4 streaming copy
+// ---------------------------------------------------------------------------
+__global__ void copy_f4(const float4* __restrict__ src,
+                        float4* __restrict__ dst, size_t n) {
+  size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = src[i];
+}
+
+double bandwidth_gbs(int device, int mib, int iters) {
+  HIP_CHECK(hipSetDevice(device));
+  size_t bytes = (size_t)mib << 20;
+  size_t n = bytes / sizeof(float4);
+  float4 *src = nullptr, *dst = nullptr;
+  HIP_CHECK(hipMalloc(&src, bytes));
+  HIP_CHECK(hipMalloc(&dst, bytes));
+  HIP_CHECK(hipMemset(src, 1, bytes));
+  dim3 block(256);
+  // >> 256 workgroups to fill 8 XCDs; cap so the tail loop stays short.
+  dim3 grid((unsigned)std::min<size_t>((n + 255) / 256, 8192));
+  // warmup
+  hipLaunchKernelGGL(copy_f4, grid, block, 0, 0, src, dst, n);
+  HIP_CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL(copy_f4, grid, block, 0, 0, src, dst, n);
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  HIP_CHECK(hipFree(src));
+  HIP_CHECK(hipFree(dst));
+  double gb = 2.0 * (double)bytes * iters / 1e9;  // read + write
+  return gb / (ms / 1e3);
+}
+
+// ---------------------------------------------------------------------------
+// MFMA: v_mfma_f32_16x16x32_bf16
+// ---------------------------------------------------------------------------
+typedef short bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr short kBf16One = 0x3F80;  // 1.0 in bf16 bit pattern
+
+__global__ void mfma_ones(float* out, int iters) {
+  bf16x8 a, b;
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    a[k] = kBf16One;
+    b[k] = kBf16One;
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < iters; ++i)
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  int lane = threadIdx.x & 63;
+  size_t base = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) out[base + r] = acc[r];
+}
+
+// Throughput burn: 8 independent accumulators so back-to-back issue is not
+// dependency-limited (~17 cyc/SIMD issue for 16x16x32).
+__global__ void __launch_bounds__(256, 4) mfma_burn(float* sink, int iters) {
+  bf16x8 a, b;
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    a[k] = kBf16One;
+    b[k] = (short)(kBf16One + (threadIdx.x & 1));
+  }
+  f32x4 acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = {0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < iters; ++i) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[j], 0, 0, 0);
+  }
+  float s = 0;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) s += acc[j][0] + acc[j][1] + acc[j][2] + acc[j][3];
+  if (s == -1.f) sink[blockIdx.x] = s;  // never true; defeats DCE
+}
+
+py::dict mfma_check(int device) {
+  HIP_CHECK(hipSetDevice(device));
+  const int iters = 4;
+  const int block = 256, grid = 64;
+  size_t n = (size_t)block * grid * 4;
+  float* out = nullptr;
+  HIP_CHECK(hipMalloc(&out, n * sizeof(float)));
+  hipLaunchKernelGGL(mfma_ones, dim3(grid), dim3(block), 0, 0, out, iters);
+  HIP_CHECK(hipDeviceSynchronize());
+  std::vector<float> host(n);
+  HIP_CHECK(hipMemcpy(host.data(), out, n * sizeof(float),
+                      hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(out));
+  // ones(16x32) x ones(32x16): every element = K = 32 per iteration.
+  const float want = 32.0f * iters;
+  size_t bad = 0;
+  for (float v : host)
+    if (v != want) ++bad;
+  py::dict d;
+  d["ok"] = (bad == 0);
+  d["expected"] = want;
+  d["mismatches"] = bad;
+  d["checked"] = n;
+  return d;
+}
+
+double mfma_tflops(int device, int iters, int blocks) {
+  HIP_CHECK(hipSetDevice(device));
+  float* sink = nullptr;
+  HIP_CHECK(hipMalloc(&sink, blocks * sizeof(float)));
+  dim3 grid(blocks), block(256);
+  hipLaunchKernelGGL(mfma_burn, grid, block, 0, 0, sink, 64);  // warmup
+  HIP_CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventRecord(t0));
+  hipLaunchKernelGGL(mfma_burn, grid, block, 0, 0, sink, iters);
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  HIP_CHECK(hipFree(sink));
+  // flops = 2*M*N*K per MFMA; 4 waves/block, 8 MFMAs per inner iter.
+  double mfmas = (double)blocks * (256 / 64) * 8 * iters;
+  double flops = mfmas * 2.0 * 16 * 16 * 32;
+  return flops / (ms / 1e3) / 1e12;
+}
+
+// ---------------------------------------------------------------------------
+// burner for sharing demos
+// ---------------------------------------------------------------------------
+__global__ void valu_burn(float* sink, long long iters) {
+  float x = 1.0f + threadIdx.x * 1e-6f;
+  for (long long i = 0; i < iters; ++i) x = fmaf(x, 1.0000001f, 1e-7f);
+  if (x == -1.f) sink[blockIdx.x] = x;
+}
+
+double burn_ms(int device, int millis) {
+  HIP_CHECK(hipSetDevice(device));
+  float* sink = nullptr;
+  HIP_CHECK(hipMalloc(&sink, 4096 * sizeof(float)));
+  // calibrate: ~2 cycles per fma at 2.4 GHz -> ~1.2e6 iters/ms
+  long long iters = (long long)millis * 1200000LL;
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventRecord(t0));
+  hipLaunchKernelGGL(valu_burn, dim3(1024), dim3(256), 0, 0, sink, iters);
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  HIP_CHECK(hipFree(sink));
+  return (double)ms;
+}
+
+// ---------------------------------------------------------------------------
+// device info
+// ---------------------------------------------------------------------------
+int device_count() {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) return 0;
+  return n;
+}
+
+py::dict device_info(int device) {
+  hipDeviceProp_t prop;
+  HIP_CHECK(hipGetDeviceProperties(&prop, device));
+  py::dict d;
+  d["name"] = std::string(prop.name);
+  d["gcn_arch"] = std::string(prop.gcnArchName);
+  d["total_mem_mib"] = (size_t)(prop.totalGlobalMem >> 20);
+  d["multi_processor_count"] = prop.multiProcessorCount;
+  d["warp_size"] = prop.warpSize;
+  d["max_threads_per_cu"] = prop.maxThreadsPerMultiProcessor;
+  return d;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_hiphealth, m) {
+  m.doc() = "gfx950 health-check and demo workload kernels";
+  m.def("device_count", &device_count);
+  m.def("device_info", &device_info, py::arg("device") = 0);
+  m.def("bandwidth_gbs", &bandwidth_gbs, py::arg("device") = 0,
+        py::arg("mib") = 1024, py::arg("iters") = 10);
+  m.def("mfma_check", &mfma_check, py::arg("device") = 0);
+  m.def("mfma_tflops", &mfma_tflops, py::arg("device") = 0,
+        py::arg("iters") = 8192, py::arg("blocks") = 2048);
+  m.def("burn_ms", &burn_ms, py::arg("device") = 0, py::arg("millis") = 100);
+}

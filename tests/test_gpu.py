@@ -1,0 +1,163 @@
+"""GPU-marked tests: run on a real MI355X via gpurun.
+
+These exercise the NATIVE path end-to-end: the _amdhal amdsmi binding, KFD
+sysfs topology, real device nodes in CDI specs, and the gfx950 HIP health
+kernels. They fail loudly (no eager/fake fallback) if the native
+extensions are missing.
+"""
+
+import os
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def real_lib():
+    from k8s_dra_driver_amd.hal.amdsmi import AmdSmiDeviceLib
+
+    lib = AmdSmiDeviceLib()
+    lib.open()
+    yield lib
+    lib.close()
+
+
+class TestNativeEnumeration:
+    def test_amdhal_binds(self):
+        from k8s_dra_driver_amd import _amdhal
+
+        v = _amdhal.lib_version()
+        assert v["major"] >= 24
+
+    def test_enumerate_real_gpu(self, real_lib):
+        gpus = real_lib.enumerate()
+        assert len(gpus) >= 1
+        g = gpus[0]
+        assert g.architecture == "gfx950"
+        assert g.vram_total_mib > 200 * 1024  # 288 GiB HBM3E
+        assert g.cu_count == 256
+        assert g.render_minor >= 128
+        assert g.uuid
+        assert g.compute_partition in ("SPX", "DPX", "TPX", "QPX", "CPX")
+        assert g.memory_partition.startswith("NPS")
+
+    def test_kfd_topology_matches_amdsmi(self, real_lib):
+        nodes = real_lib.topology.gpu_nodes()
+        assert len(nodes) >= 1
+        assert nodes[0].gfx_arch == "gfx950"
+        minors = {n.render_minor for n in nodes}
+        for g in real_lib.enumerate():
+            assert g.render_minor in minors
+
+    def test_device_nodes_exist(self, real_lib):
+        paths = real_lib.device_node_paths(0)
+        assert os.path.exists(paths["kfd"])
+        assert os.path.exists(paths["renderD"])
+
+    def test_health_check(self, real_lib):
+        h = real_lib.health_check(0)
+        assert h["status"] == "healthy"
+
+
+class TestPreparePathOnHardware:
+    def test_full_prepare_with_real_hal(self, real_lib, tmp_path):
+        from k8s_dra_driver_amd.cdi.spec import read_spec_file
+        from k8s_dra_driver_amd.kube.client import InMemoryKube
+        from k8s_dra_driver_amd.plugin.driver import ClaimRef, Driver
+
+        kube = InMemoryKube()
+        driver = Driver(
+            real_lib,
+            kube,
+            node_name="gpu-node",
+            cdi_root=str(tmp_path / "cdi"),
+            checkpoint_root=str(tmp_path / "state"),
+            use_tmpfs=False,
+        )
+        driver.startup()
+        slices = kube.list_resource_slices("gpu.amd.com")
+        assert slices and slices[0]["spec"]["devices"]
+        dev_name = slices[0]["spec"]["devices"][0]["name"]
+
+        kube.put_resource_claim(
+            {
+                "metadata": {
+                    "namespace": "default",
+                    "name": "hw-claim",
+                    "uid": "hw-uid",
+                },
+                "status": {
+                    "allocation": {
+                        "devices": {
+                            "results": [
+                                {
+                                    "request": "gpu",
+                                    "driver": "gpu.amd.com",
+                                    "pool": "gpu-node",
+                                    "device": dev_name,
+                                }
+                            ]
+                        }
+                    }
+                },
+            }
+        )
+        res = driver.node_prepare_resources(
+            [ClaimRef("default", "hw-claim", "hw-uid")]
+        )["hw-uid"]
+        assert not res.error
+        base = read_spec_file(str(tmp_path / "cdi" / "k8s.gpu.amd.com-device.json"))
+        node_paths = [
+            n["path"]
+            for d in base["devices"]
+            for n in d["containerEdits"]["deviceNodes"]
+        ]
+        for p in node_paths:
+            assert os.path.exists(p), f"CDI references missing node {p}"
+        driver.node_unprepare_resources(
+            [ClaimRef("default", "hw-claim", "hw-uid")]
+        )
+
+
+class TestHipHealth:
+    def test_device_info(self):
+        from k8s_dra_driver_amd import _hiphealth
+
+        assert _hiphealth.device_count() >= 1
+        info = _hiphealth.device_info(0)
+        assert "gfx950" in info["gcn_arch"]
+        assert info["warp_size"] == 64
+        assert info["multi_processor_count"] == 256
+
+    def test_mfma_correctness(self):
+        from k8s_dra_driver_amd import _hiphealth
+
+        check = _hiphealth.mfma_check(0)
+        assert check["ok"], check
+
+    def test_bandwidth_sane(self):
+        from k8s_dra_driver_amd import _hiphealth
+
+        bw = _hiphealth.bandwidth_gbs(0, 1024, 10)
+        # HBM3E: 8 TB/s peak, ~6.3 achievable; anything under 2 TB/s means
+        # the copy never hit the real device.
+        assert bw > 2000, f"bandwidth {bw} GB/s too low"
+
+    def test_mfma_throughput_sane(self):
+        from k8s_dra_driver_amd import _hiphealth
+
+        tf = _hiphealth.mfma_tflops(0, 2048, 2048)
+        # dense bf16 peak ~2.5 PF; a health kernel should clear 0.5 PF
+        assert tf > 500, f"MFMA throughput {tf} TF/s too low"
+
+
+class TestTorchInterop:
+    def test_torch_sees_gpu(self):
+        import torch
+
+        assert torch.cuda.is_available()
+        x = torch.randn(256, 256, device="cuda:0", requires_grad=True)
+        (x @ x).sum().backward()
+        torch.cuda.synchronize()
+        assert x.grad is not None

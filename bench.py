@@ -1,0 +1,251 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GPU-pod scheduling throughput + claim latency.
+
+Measures the project's north-star metric (BASELINE.json): ResourceClaim
+allocation/prepare p50 latency and pods-with-GPU scheduled per second on an
+MI355X node, through the FULL driver pipeline:
+
+    synthetic ResourceClaim -> allocator assigns a published device ->
+    gRPC NodePrepareResources over the plugin's unix socket ->
+    (config decode, sharing, CDI spec write, checkpoint write) ->
+    gRPC NodeUnprepareResources
+
+One rank per GPU (torchrun for N>1); each rank runs its own plugin
+instance scoped to its GPU and drives `--pods-per-step` pod lifecycles per
+step. Weak scaling: per-GPU work is fixed as N grows.
+
+The HAL is the real amdsmi backend when a GPU is present (enumeration,
+render minors, CDI all real); otherwise the fake 8xMI355X backend
+(identical driver code path; reported in config.hal). The reference
+publishes no numbers for this metric -> vs_baseline null.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import tempfile
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import grpc
+
+from k8s_dra_driver_amd import DRIVER_NAME
+from k8s_dra_driver_amd.hal.base import HalUnavailable
+from k8s_dra_driver_amd.kube.client import InMemoryKube
+from k8s_dra_driver_amd.plugin.driver import Driver
+from k8s_dra_driver_amd.plugin.proto import V1BETA1
+from k8s_dra_driver_amd.plugin.server import PluginServer
+
+
+def pick_hal():
+    try:
+        from k8s_dra_driver_amd.hal.amdsmi import AmdSmiDeviceLib
+
+        lib = AmdSmiDeviceLib()
+        lib.open()
+        if lib.enumerate():
+            return lib, "amdsmi"
+        lib.close()
+    except (HalUnavailable, Exception):
+        pass
+    from k8s_dra_driver_amd.hal import FakeDeviceLib
+
+    lib = FakeDeviceLib()
+    lib.open()
+    return lib, "fake"
+
+
+def make_claim_obj(uid: str, node: str, device: str) -> dict:
+    return {
+        "metadata": {"namespace": "default", "name": f"claim-{uid}", "uid": uid},
+        "status": {
+            "allocation": {
+                "devices": {
+                    "results": [
+                        {
+                            "request": "gpu",
+                            "driver": DRIVER_NAME,
+                            "pool": node,
+                            "device": device,
+                        }
+                    ]
+                }
+            }
+        },
+    }
+
+
+class BenchRank:
+    """One rank's plugin stack + claim generator."""
+
+    def __init__(self, rank: int, target_gpu: int):
+        self.rank = rank
+        self.tmp = tempfile.mkdtemp(prefix=f"dra-bench-r{rank}-")
+        self.node = f"bench-node-{rank}"
+        self.lib, self.hal_kind = pick_hal()
+        self.kube = InMemoryKube()
+        self.driver = Driver(
+            self.lib,
+            self.kube,
+            node_name=self.node,
+            cdi_root=os.path.join(self.tmp, "cdi"),
+            checkpoint_root=os.path.join(self.tmp, "state"),
+            use_tmpfs=False,
+        )
+        self.driver.startup()
+        # allocate against the published slice for this rank's GPU
+        slices = self.kube.list_resource_slices(DRIVER_NAME)
+        devices = [d["name"] for s in slices for d in s["spec"]["devices"]]
+        if not devices:
+            raise RuntimeError("no devices published")
+        self.device = devices[target_gpu % len(devices)]
+
+        self.server = PluginServer(
+            self.driver, plugin_dir=os.path.join(self.tmp, "plugin")
+        )
+        self.server.start()
+        self.channel = grpc.insecure_channel(f"unix://{self.server.plugin_sock}")
+        m = V1BETA1
+        self.prepare = self.channel.unary_unary(
+            f"/{m.service_name}/NodePrepareResources",
+            request_serializer=lambda x: x.SerializeToString(),
+            response_deserializer=m.NodePrepareResourcesResponse.FromString,
+        )
+        self.unprepare = self.channel.unary_unary(
+            f"/{m.service_name}/NodeUnprepareResources",
+            request_serializer=lambda x: x.SerializeToString(),
+            response_deserializer=m.NodeUnprepareResourcesResponse.FromString,
+        )
+        self._seq = 0
+        self.latencies_ms: list = []
+
+    def step(self, pods: int) -> None:
+        """One step = `pods` full pod lifecycles (prepare + unprepare),
+        issued one at a time like kubelet does per pod admission."""
+        m = V1BETA1
+        for _ in range(pods):
+            self._seq += 1
+            uid = f"r{self.rank}-{self._seq}"
+            self.kube.put_resource_claim(
+                make_claim_obj(uid, self.node, self.device)
+            )
+            req = m.NodePrepareResourcesRequest()
+            c = req.claims.add()
+            c.namespace, c.name, c.uid = "default", f"claim-{uid}", uid
+            t0 = time.perf_counter()
+            resp = self.prepare(req)
+            dt = (time.perf_counter() - t0) * 1e3
+            err = resp.claims[uid].error
+            if err:
+                raise RuntimeError(f"prepare failed: {err}")
+            self.latencies_ms.append(dt)
+            ureq = m.NodeUnprepareResourcesRequest()
+            uc = ureq.claims.add()
+            uc.namespace, uc.name, uc.uid = "default", f"claim-{uid}", uid
+            uresp = self.unprepare(ureq)
+            if uresp.claims[uid].error:
+                raise RuntimeError(f"unprepare failed: {uresp.claims[uid].error}")
+
+    def close(self):
+        self.channel.close()
+        self.server.stop()
+        self.lib.close()
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--pods-per-step", type=int, default=32)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    import torch
+    import torch.distributed as dist
+
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group(
+            backend="nccl" if use_cuda else "gloo", rank=rank, world_size=world
+        )
+
+    bench = BenchRank(rank, target_gpu=local_rank)
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        if distributed:
+            dist.barrier()
+            if use_cuda:
+                torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        bench.step(args.pods_per_step)
+    bench.latencies_ms.clear()
+
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        bench.step(args.pods_per_step)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks of the timed region
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if use_cuda:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    lat = sorted(bench.latencies_ms)
+    p50 = statistics.median(lat) if lat else 0.0
+    p99 = lat[int(len(lat) * 0.99) - 1] if len(lat) >= 2 else p50
+    pods_total = world * args.pods_per_step * args.steps
+    result = {
+        "metric": "gpu_pods_scheduled_per_sec",
+        "value": round(pods_total / elapsed, 2),
+        "unit": "pods/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "n/a",
+        "data": "synthetic ResourceClaims (no model weights), random device attrs from HAL",
+        "config": {
+            "model": "dra-claim-lifecycle",
+            "global_batch": pods_total,
+            "seq_len": 0,
+            "parallelism": f"plugin-per-gpu x{world}",
+            "pods_per_step": args.pods_per_step,
+            "hal": bench.hal_kind,
+            "prepare_p50_ms": round(p50, 3),
+            "prepare_p99_ms": round(p99, 3),
+            "grpc": "unix-socket v1beta1",
+        },
+    }
+    bench.close()
+    if distributed:
+        dist.destroy_process_group()
+    if rank == 0:
+        print(json.dumps(result))
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -83,15 +83,18 @@ class CDIHandler:
                 else "",
             )
         ]
+        # card nodes are optional: some deployments expose only renderD*
+        # (compute needs kfd + renderD only); referencing a missing node
+        # would fail container creation.
         if card_minor >= 0:
-            nodes.append(
-                DeviceNode(
-                    path=f"/dev/dri/card{card_minor}",
-                    host_path=f"{self.dev_root}/dev/dri/card{card_minor}"
-                    if self.dev_root
-                    else "",
+            card_host = f"{self.dev_root}/dev/dri/card{card_minor}"
+            if os.path.exists(card_host):
+                nodes.append(
+                    DeviceNode(
+                        path=f"/dev/dri/card{card_minor}",
+                        host_path=card_host if self.dev_root else "",
+                    )
                 )
-            )
         return nodes
 
     # -- base spec -----------------------------------------------------------

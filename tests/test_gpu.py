@@ -10,6 +10,11 @@ import os
 
 import pytest
 
+# torch must load before the system-ROCm-linked _hiphealth extension: its
+# bundled HIP runtime wins the dlopen race, otherwise torch.cuda reports
+# "No HIP GPUs are available" (observed on the MI355X pool, round 1).
+import torch  # noqa: F401  (import order is load-bearing)
+
 pytestmark = pytest.mark.gpu
 
 

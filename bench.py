@@ -35,6 +35,7 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import grpc
 
 from k8s_dra_driver_amd import DRIVER_NAME
+from k8s_dra_driver_amd.allocator.structured import Allocator
 from k8s_dra_driver_amd.hal.base import HalUnavailable
 from k8s_dra_driver_amd.kube.client import InMemoryKube
 from k8s_dra_driver_amd.plugin.driver import Driver
@@ -60,21 +61,15 @@ def pick_hal():
     return lib, "fake"
 
 
-def make_claim_obj(uid: str, node: str, device: str) -> dict:
+def make_claim_spec(uid: str) -> dict:
+    """Synthetic unallocated ResourceClaim (what a pod template stamps)."""
     return {
         "metadata": {"namespace": "default", "name": f"claim-{uid}", "uid": uid},
-        "status": {
-            "allocation": {
-                "devices": {
-                    "results": [
-                        {
-                            "request": "gpu",
-                            "driver": DRIVER_NAME,
-                            "pool": node,
-                            "device": device,
-                        }
-                    ]
-                }
+        "spec": {
+            "devices": {
+                "requests": [
+                    {"name": "gpu", "deviceClassName": "any.gpu.amd.com", "count": 1}
+                ]
             }
         },
     }
@@ -98,12 +93,14 @@ class BenchRank:
             use_tmpfs=False,
         )
         self.driver.startup()
-        # allocate against the published slice for this rank's GPU
+        # this rank schedules onto its own GPU's published devices
         slices = self.kube.list_resource_slices(DRIVER_NAME)
-        devices = [d["name"] for s in slices for d in s["spec"]["devices"]]
-        if not devices:
+        all_devices = [d for s in slices for d in s["spec"]["devices"]]
+        if not all_devices:
             raise RuntimeError("no devices published")
-        self.device = devices[target_gpu % len(devices)]
+        mine = all_devices[target_gpu % len(all_devices)]
+        self.devices = [mine]
+        self.allocator = Allocator()
 
         self.server = PluginServer(
             self.driver, plugin_dir=os.path.join(self.tmp, "plugin")
@@ -125,19 +122,23 @@ class BenchRank:
         self.latencies_ms: list = []
 
     def step(self, pods: int) -> None:
-        """One step = `pods` full pod lifecycles (prepare + unprepare),
-        issued one at a time like kubelet does per pod admission."""
+        """One step = `pods` full pod lifecycles, each doing the whole
+        pipeline: allocation (CEL + topology scoring) -> apiserver write ->
+        gRPC prepare -> gRPC unprepare. Issued one at a time like kubelet
+        does per pod admission."""
         m = V1BETA1
         for _ in range(pods):
             self._seq += 1
             uid = f"r{self.rank}-{self._seq}"
-            self.kube.put_resource_claim(
-                make_claim_obj(uid, self.node, self.device)
+            t0 = time.perf_counter()  # pod-sees-GPU latency starts here
+            claim = make_claim_spec(uid)
+            self.allocator.allocate_into_claim(
+                claim, self.devices, pool=self.node, node_name=self.node
             )
+            self.kube.put_resource_claim(claim)
             req = m.NodePrepareResourcesRequest()
             c = req.claims.add()
             c.namespace, c.name, c.uid = "default", f"claim-{uid}", uid
-            t0 = time.perf_counter()
             resp = self.prepare(req)
             dt = (time.perf_counter() - t0) * 1e3
             err = resp.claims[uid].error
@@ -234,9 +235,9 @@ def main() -> int:
             "parallelism": f"plugin-per-gpu x{world}",
             "pods_per_step": args.pods_per_step,
             "hal": bench.hal_kind,
-            "prepare_p50_ms": round(p50, 3),
-            "prepare_p99_ms": round(p99, 3),
-            "grpc": "unix-socket v1beta1",
+            "alloc_prepare_p50_ms": round(p50, 3),
+            "alloc_prepare_p99_ms": round(p99, 3),
+            "pipeline": "CEL-alloc + grpc prepare + cdi + checkpoint, unix-socket v1beta1",
         },
     }
     bench.close()

@@ -1,0 +1,75 @@
+"""Cluster controller CLI (reference cmd/nvidia-dra-controller/main.go)."""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import signal
+import threading
+
+from ..metrics.prom import PluginMetrics
+
+log = logging.getLogger(__name__)
+
+
+def _env(name: str, default: str = "") -> str:
+    return os.environ.get(name, default)
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser("amd-dra-controller")
+    p.add_argument("--kubeconfig", default=_env("KUBECONFIG", ""))
+    p.add_argument(
+        "--poll-interval", type=float, default=float(_env("POLL_INTERVAL", "10"))
+    )
+    p.add_argument(
+        "--metrics-port", type=int, default=int(_env("METRICS_PORT", "8085"))
+    )
+    p.add_argument(
+        "--manage-node-labels",
+        default=_env("MANAGE_NODE_LABELS", "true").lower() == "true",
+        action="store_true",
+    )
+    p.add_argument("-v", "--verbosity", type=int, default=int(_env("LOG_LEVEL", "1")))
+    return p
+
+
+def main(argv=None) -> int:
+    args = build_parser().parse_args(argv)
+    logging.basicConfig(
+        level=logging.DEBUG if args.verbosity >= 4 else logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s %(message)s",
+    )
+    if args.kubeconfig == "memory":
+        from ..kube.client import InMemoryKube
+
+        kube = InMemoryKube()
+    else:
+        from ..kube.http_kube import HttpKube
+
+        kube = HttpKube(kubeconfig=args.kubeconfig or None)
+
+    from .manager import ControllerManager
+
+    metrics = PluginMetrics()
+    if args.metrics_port:
+        metrics.serve(args.metrics_port)
+
+    mgr = ControllerManager(
+        kube,
+        poll_interval=args.poll_interval,
+        manage_labels=args.manage_node_labels,
+    )
+    mgr.start()
+    stop = threading.Event()
+    for sig in (signal.SIGTERM, signal.SIGINT):
+        signal.signal(sig, lambda *_: stop.set())
+    log.info("amd-dra-controller ready")
+    stop.wait()
+    mgr.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -152,6 +152,41 @@ class Driver:
             self.metrics.prepared_claims.inc()
             return ClaimResult(devices)
 
+    # ------------------------------------------------------------------
+    def cleanup_orphans(self) -> List[str]:
+        """Unprepare claims whose ResourceClaim object is gone or has been
+        re-created with a new UID — the cleanup the reference leaves as a
+        TODO (driver.go:156-168: leaked CDI files / sharing artifacts for
+        deleted claims). Returns the UIDs cleaned."""
+        cleaned: List[str] = []
+        for uid, pc in self.state.checkpoints.list_all().items():
+            stale = False
+            if not pc.namespace or not pc.name:
+                continue
+            try:
+                cur = self.kube.get_resource_claim(pc.namespace, pc.name)
+                if (cur.get("metadata") or {}).get("uid") != uid:
+                    stale = True
+            except NotFound:
+                stale = True
+            except Exception:
+                continue  # apiserver hiccup: never GC on uncertainty
+            if stale:
+                try:
+                    self.state.unprepare(uid)
+                    cleaned.append(uid)
+                except Exception:
+                    log.exception("orphan cleanup of %s failed", uid)
+        # claim CDI specs with no checkpoint (crash between the two writes)
+        live = set(self.state.checkpoints.list_all())
+        for uid in self.state.cdi.list_claim_spec_uids():
+            if uid not in live and uid not in cleaned:
+                self.state.cdi.delete_claim_spec(uid)
+                cleaned.append(uid)
+        if cleaned:
+            log.info("cleaned %d orphaned claim(s): %s", len(cleaned), cleaned[:5])
+        return cleaned
+
     def node_unprepare_resources(
         self, claims: List[ClaimRef]
     ) -> Dict[str, ClaimResult]:

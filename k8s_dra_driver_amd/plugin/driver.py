@@ -87,6 +87,13 @@ class Driver:
         )
         # Republish whenever the allocatable set changes (repartition).
         self.state.on_allocatable_change = self.publish_resources
+        # Failure detection: unhealthy GPUs are pulled from publication
+        # (start()ed by main.py; tests drive check_once directly).
+        from .health import HealthMonitor
+
+        self.health = HealthMonitor(
+            lib, on_change=lambda _unhealthy: self.publish_resources()
+        )
 
     # ------------------------------------------------------------------
     def startup(self) -> None:
@@ -104,7 +111,12 @@ class Driver:
                 log.exception("unpublish failed")
 
     def publish_resources(self) -> None:
-        devices = [d.to_device() for d in self.state.allocatable_devices()]
+        unhealthy = self.health.unhealthy_gpus
+        devices = [
+            d.to_device()
+            for d in self.state.allocatable_devices()
+            if d.parent_gpu.index not in unhealthy
+        ]
         self.publisher.publish(devices)
         self.metrics.allocatable_devices.set(len(devices))
 

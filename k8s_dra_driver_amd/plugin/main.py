@@ -111,6 +111,7 @@ def main(argv=None) -> int:
         metrics=metrics,
     )
     driver.startup()
+    driver.health.start()  # failure detection -> slice self-healing
     server = PluginServer(
         driver,
         plugin_dir=args.plugin_path,
@@ -118,13 +119,24 @@ def main(argv=None) -> int:
     )
     server.start()
 
+    # periodic orphan cleanup (reference TODO parity, driver.go:156-168)
     stop = threading.Event()
+
+    def cleanup_loop():
+        while not stop.wait(300):
+            try:
+                driver.cleanup_orphans()
+            except Exception:
+                log.exception("orphan cleanup pass failed")
+
+    threading.Thread(target=cleanup_loop, name="orphan-gc", daemon=True).start()
     for sig in (signal.SIGTERM, signal.SIGINT):
         signal.signal(sig, lambda *_: stop.set())
     log.info(
         "amd-dra-kubeletplugin ready: node=%s hal=%s", args.node_name, args.hal
     )
     stop.wait()
+    driver.health.stop()
     server.stop()
     driver.shutdown()
     lib.close()

@@ -1,0 +1,55 @@
+"""Health monitor tests: fault-injected failure detection + slice healing."""
+
+from k8s_dra_driver_amd import DRIVER_NAME
+from k8s_dra_driver_amd.hal import FakeDeviceLib
+from k8s_dra_driver_amd.hal.base import HalError
+from k8s_dra_driver_amd.kube.client import InMemoryKube
+from k8s_dra_driver_amd.plugin.driver import Driver
+from k8s_dra_driver_amd.plugin.health import HealthMonitor
+
+
+def test_hysteresis_and_recovery(fake_lib):
+    events = []
+    mon = HealthMonitor(
+        fake_lib,
+        on_change=lambda s: events.append(set(s)),
+        failures_to_unhealthy=2,
+    )
+    # one failure: below threshold
+    fake_lib.faults.fail_next("health_check", HalError("hw error"))
+    assert mon.check_once() == set()
+    # second consecutive failure on gpu-0 -> unhealthy
+    fake_lib.faults.fail_next("health_check", HalError("hw error"))
+    assert mon.check_once() == {0}
+    assert events == [{0}]
+    # healthy pass -> recovered
+    assert mon.check_once() == set()
+    assert events == [{0}, set()]
+
+
+def test_unhealthy_gpu_pulled_from_slices(tmp_path):
+    lib = FakeDeviceLib()
+    lib.open()
+    kube = InMemoryKube()
+    driver = Driver(
+        lib,
+        kube,
+        node_name="n",
+        cdi_root=str(tmp_path / "cdi"),
+        checkpoint_root=str(tmp_path / "state"),
+        use_tmpfs=False,
+    )
+    driver.startup()
+    assert (
+        len(kube.list_resource_slices(DRIVER_NAME)[0]["spec"]["devices"]) == 8
+    )
+    # gpu-0 fails in two consecutive polls (the injector pops one error
+    # per call; the first health_check of each round is gpu-0)
+    driver.health.failures_to_unhealthy = 2
+    lib.faults.fail_next("health_check", HalError("dead"))
+    driver.health.check_once()
+    lib.faults.fail_next("health_check", HalError("dead"))
+    driver.health.check_once()
+    devices = kube.list_resource_slices(DRIVER_NAME)[0]["spec"]["devices"]
+    assert len(devices) == 7
+    assert "gpu-0" not in [d["name"] for d in devices]

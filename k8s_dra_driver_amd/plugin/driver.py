@@ -127,6 +127,8 @@ class Driver:
         """Batch prepare; one result per claim UID."""
         if not claims:
             return {}
+        if len(claims) == 1:  # kubelet's common case: skip pool dispatch
+            return {claims[0].uid: self._prepare_one(claims[0])}
         futures = {
             c.uid: self._pool.submit(self._prepare_one, c) for c in claims
         }
@@ -202,6 +204,8 @@ class Driver:
     def node_unprepare_resources(
         self, claims: List[ClaimRef]
     ) -> Dict[str, ClaimResult]:
+        if len(claims) == 1:
+            return {claims[0].uid: self._unprepare_one(claims[0])}
         futures = {
             c.uid: self._pool.submit(self._unprepare_one, c) for c in claims
         }

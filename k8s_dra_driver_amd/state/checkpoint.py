@@ -55,6 +55,10 @@ class PreparedClaim:
     shared_session_id: str = ""
     #: partition bookkeeping: gpu_index -> (compute, memory) applied
     repartitioned: Dict[str, List[str]] = field(default_factory=dict)
+    #: claim-level CDI edits, persisted so the claim spec file can be
+    #: regenerated if a crash lands between the two hot-path writes
+    claim_env: List[str] = field(default_factory=list)
+    claim_mounts: List[dict] = field(default_factory=list)
 
     def to_v1(self) -> dict:
         return asdict(self)

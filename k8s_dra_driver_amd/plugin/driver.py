@@ -16,6 +16,7 @@ errors, exactly like the reference (``driver.go:96-116``).
 from __future__ import annotations
 
 import logging
+import time
 from concurrent.futures import ThreadPoolExecutor
 from dataclasses import dataclass
 from typing import Dict, List, Optional
@@ -155,7 +156,16 @@ class Driver:
                     f"have {got_uid}, prepare was for {ref.uid}",
                 )
             try:
+                t0 = time.perf_counter()
                 devices = self.state.prepare(claim)
+                log.debug(
+                    "prepared %s/%s uid=%s devices=%d in %.2fms",
+                    ref.namespace,
+                    ref.name,
+                    ref.uid,
+                    len(devices),
+                    (time.perf_counter() - t0) * 1e3,
+                )
             except PrepareError as e:
                 self.metrics.prepare_errors.inc()
                 return ClaimResult([], str(e))

@@ -54,6 +54,10 @@ MEMORY_MODES: Dict[str, int] = {
 #: A compute partition may never span multiple memory domains, i.e.
 #: num_domains must divide num_partitions (or equal 1). Live hardware caps
 #: (amdsmi_get_gpu_memory_partition_config nps_cap_mask) override this.
+#: Measured round 1: the MI355X pool SKU reports compute caps
+#: SPX/DPX/QPX/CPX but memory caps NPS1,NPS2 only (amd-smi partition) —
+#: the MI300X-style NPS4 rows below are architectural upper bounds that
+#: the live-caps override prunes on such firmware.
 DEFAULT_VALID_NPS: Dict[str, Tuple[str, ...]] = {
     "SPX": ("NPS1",),
     "DPX": ("NPS1", "NPS2"),

@@ -32,15 +32,20 @@ def build_parser() -> argparse.ArgumentParser:
         action="store_true",
     )
     p.add_argument("-v", "--verbosity", type=int, default=int(_env("LOG_LEVEL", "1")))
+    p.add_argument(
+        "--logging-format",
+        default=_env("LOGGING_FORMAT", "text"),
+        choices=["text", "json"],
+        help="log output format (logsapi parity)",
+    )
     return p
 
 
 def main(argv=None) -> int:
     args = build_parser().parse_args(argv)
-    logging.basicConfig(
-        level=logging.DEBUG if args.verbosity >= 4 else logging.INFO,
-        format="%(asctime)s %(levelname)s %(name)s %(message)s",
-    )
+    from ..utils.log import setup_logging
+
+    setup_logging(args.verbosity, json_format=args.logging_format == "json")
     if args.kubeconfig == "memory":
         from ..kube.client import InMemoryKube
 

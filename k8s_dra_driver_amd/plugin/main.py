@@ -75,6 +75,12 @@ def build_parser() -> argparse.ArgumentParser:
         help="serve Prometheus /metrics on this port (0 = disabled)",
     )
     p.add_argument("-v", "--verbosity", type=int, default=int(_env("LOG_LEVEL", "1")))
+    p.add_argument(
+        "--logging-format",
+        default=_env("LOGGING_FORMAT", "text"),
+        choices=["text", "json"],
+        help="log output format (logsapi parity)",
+    )
     return p
 
 
@@ -92,10 +98,9 @@ def make_kube_client(args):
 
 def main(argv=None) -> int:
     args = build_parser().parse_args(argv)
-    logging.basicConfig(
-        level=logging.DEBUG if args.verbosity >= 4 else logging.INFO,
-        format="%(asctime)s %(levelname)s %(name)s %(message)s",
-    )
+    from ..utils.log import setup_logging
+
+    setup_logging(args.verbosity, json_format=args.logging_format == "json")
     lib = new_device_lib(args.hal)
     lib.open()
     kube = make_kube_client(args)

@@ -27,6 +27,13 @@ def build_parser() -> argparse.ArgumentParser:
         "--metrics-port", type=int, default=int(_env("METRICS_PORT", "8085"))
     )
     p.add_argument(
+        "--allocate-claims",
+        default=_env("ALLOCATE_CLAIMS", "false").lower() == "true",
+        action="store_true",
+        help="allocate pending ResourceClaims with the topology-aware "
+        "allocator (leave off when kube-scheduler performs DRA allocation)",
+    )
+    p.add_argument(
         "--manage-node-labels",
         default=_env("MANAGE_NODE_LABELS", "true").lower() == "true",
         action="store_true",
@@ -65,6 +72,7 @@ def main(argv=None) -> int:
         kube,
         poll_interval=args.poll_interval,
         manage_labels=args.manage_node_labels,
+        allocate_claims=args.allocate_claims,
     )
     mgr.start()
     stop = threading.Event()

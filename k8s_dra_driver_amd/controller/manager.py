@@ -83,11 +83,17 @@ class ControllerManager:
         poll_interval: float = 10.0,
         manage_labels: bool = True,
         cleanup_orphans: bool = True,
+        allocate_claims: bool = False,
     ):
         self.kube = kube
         self.poll_interval = poll_interval
         self.manage_labels = manage_labels
         self.cleanup_orphans = cleanup_orphans
+        self.scheduler = None
+        if allocate_claims:
+            from .scheduler import ClaimScheduler
+
+            self.scheduler = ClaimScheduler(kube)
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self._owned_labels: Dict[str, Dict[str, str]] = {}
@@ -95,6 +101,8 @@ class ControllerManager:
     # ------------------------------------------------------------------
     def reconcile_once(self) -> Dict[str, Dict[str, str]]:
         """One level-triggered pass. Returns applied labels per node."""
+        if self.scheduler is not None:
+            self.scheduler.reconcile_once()
         slices = self.kube.list_resource_slices(DRIVER_NAME)
         by_node: Dict[str, List[dict]] = {}
         for s in slices:

@@ -30,6 +30,18 @@ class KubeClient(abc.ABC):
     @abc.abstractmethod
     def get_resource_claim(self, namespace: str, name: str) -> dict: ...
 
+    def list_resource_claims(self) -> List[dict]:
+        """All ResourceClaims (controller allocator input); optional."""
+        raise NotImplementedError
+
+    def update_resource_claim_status(self, obj: dict) -> dict:
+        """Write claim.status (allocation); optional."""
+        raise NotImplementedError
+
+    def get_device_classes(self) -> List[dict]:
+        """DeviceClass objects (controller allocator input); optional."""
+        return []
+
     @abc.abstractmethod
     def create_resource_slice(self, obj: dict) -> dict: ...
 
@@ -59,6 +71,7 @@ class InMemoryKube(KubeClient):
         self.resource_claims: Dict[str, dict] = {}  # key: ns/name
         self.resource_slices: Dict[str, dict] = {}
         self.nodes: Dict[str, dict] = {}
+        self.device_classes: Dict[str, dict] = {}
         self._rv = 0
         #: watch hooks: fn(kind, verb, obj)
         self.watchers: List[Callable[[str, str, dict], None]] = []
@@ -93,6 +106,30 @@ class InMemoryKube(KubeClient):
             if key not in self.resource_claims:
                 raise NotFound(f"resourceclaim {key}")
             return copy.deepcopy(self.resource_claims[key])
+
+    def list_resource_claims(self) -> List[dict]:
+        with self._lock:
+            return [copy.deepcopy(c) for c in self.resource_claims.values()]
+
+    def update_resource_claim_status(self, obj: dict) -> dict:
+        with self._lock:
+            key = f"{obj['metadata'].get('namespace','default')}/{obj['metadata']['name']}"
+            if key not in self.resource_claims:
+                raise NotFound(f"resourceclaim {key}")
+            cur = self.resource_claims[key]
+            cur["status"] = copy.deepcopy(obj.get("status") or {})
+            cur["metadata"]["resourceVersion"] = self._next_rv()
+            self._emit("ResourceClaim", "status", cur)
+            return copy.deepcopy(cur)
+
+    def get_device_classes(self) -> List[dict]:
+        with self._lock:
+            return [copy.deepcopy(d) for d in self.device_classes.values()]
+
+    def put_device_class(self, obj: dict) -> dict:
+        with self._lock:
+            self.device_classes[obj["metadata"]["name"]] = copy.deepcopy(obj)
+            return obj
 
     def create_resource_slice(self, obj: dict) -> dict:
         with self._lock:

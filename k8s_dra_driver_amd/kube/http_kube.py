@@ -145,6 +145,23 @@ class HttpKube(KubeClient):
         out = self._req("GET", f"{RESOURCE_V1BETA1}/resourceslices{params}")
         return out.get("items", [])
 
+    def list_resource_claims(self) -> List[dict]:
+        out = self._req("GET", f"{RESOURCE_V1BETA1}/resourceclaims")
+        return out.get("items", [])
+
+    def update_resource_claim_status(self, obj: dict) -> dict:
+        ns = obj["metadata"].get("namespace", "default")
+        name = obj["metadata"]["name"]
+        return self._req(
+            "PUT",
+            f"{RESOURCE_V1BETA1}/namespaces/{ns}/resourceclaims/{name}/status",
+            obj,
+        )
+
+    def get_device_classes(self) -> List[dict]:
+        out = self._req("GET", f"{RESOURCE_V1BETA1}/deviceclasses")
+        return out.get("items", [])
+
     def get_node(self, name: str) -> dict:
         return self._req("GET", f"{CORE_V1}/nodes/{name}")
 

@@ -1,0 +1,149 @@
+"""amd-dra-ctl — operational CLI for the driver's hardware layer.
+
+The nvidia-smi-adjacent workflows an operator needs around this driver:
+inspect devices/topology as the driver sees them, dump the ResourceSlice
+projection, carve/restore partitions, and run the health probes.
+
+    amd-dra-ctl list                 # devices incl. partitions
+    amd-dra-ctl topology             # xGMI adjacency matrix
+    amd-dra-ctl slice                # ResourceSlice device JSON
+    amd-dra-ctl partition 0 CPX NPS1 # dynamic repartition (needs idle GPU)
+    amd-dra-ctl health               # per-GPU health + HIP probes
+Use ``--hal fake`` anywhere to drive the modeled 8xMI355X node.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+
+from .hal import new_device_lib
+from .hal.model import AllocatableDevice
+
+
+def _lib(args):
+    lib = new_device_lib(args.hal)
+    lib.open()
+    return lib
+
+
+def cmd_list(args) -> int:
+    lib = _lib(args)
+    for g in lib.enumerate():
+        print(
+            f"{g.canonical_name}: {g.product_name} [{g.architecture}] "
+            f"uuid={g.uuid} oam={g.oam_id} vram={g.vram_total_mib}MiB "
+            f"cu={g.cu_count} renderD{g.render_minor} "
+            f"mode={g.compute_partition}/{g.memory_partition}"
+        )
+        for p in g.partitions:
+            print(
+                f"  {p.canonical_name}: kfd={p.kfd_node_id} "
+                f"renderD{p.render_minor} "
+                f"cu={p.profile.cus_per_partition} "
+                f"mem={p.profile.memory_mib_per_partition}MiB "
+                f"domain={p.profile.memory_domain_of(p.partition_id)}"
+            )
+    return 0
+
+
+def cmd_topology(args) -> int:
+    lib = _lib(args)
+    gpus = lib.enumerate()
+    ids = [g.oam_id for g in gpus]
+    print("xGMI adjacency (links between OAM ids):")
+    print("     " + " ".join(f"{i:>3}" for i in ids))
+    for g in gpus:
+        peers = set(g.xgmi_peer_oam_ids())
+        row = " ".join(
+            "  x" if i == g.oam_id else ("  1" if i in peers else "  .")
+            for i in ids
+        )
+        print(f"{g.oam_id:>4} {row}")
+    hives = {g.xgmi_hive_id for g in gpus}
+    print(f"hives: {sorted(hives)}")
+    return 0
+
+
+def cmd_slice(args) -> int:
+    lib = _lib(args)
+    devices = []
+    for g in lib.enumerate():
+        if g.partitions:
+            devices.extend(
+                AllocatableDevice.from_partition(g, p).to_device()
+                for p in g.partitions
+            )
+        else:
+            devices.append(AllocatableDevice.from_gpu(g).to_device())
+    print(json.dumps(devices, indent=2, sort_keys=True))
+    return 0
+
+
+def cmd_partition(args) -> int:
+    from .partition.manager import PartitionManager, RepartitionRefused
+
+    lib = _lib(args)
+    mgr = PartitionManager(lib)
+    try:
+        switched = mgr.ensure_mode(
+            args.gpu, args.compute.upper(), args.memory.upper(), allow_dynamic=True
+        )
+    except (RepartitionRefused, ValueError) as e:
+        print(f"refused: {e}", file=sys.stderr)
+        return 1
+    g = lib.enumerate()[args.gpu]
+    print(
+        f"gpu-{args.gpu}: {'switched to' if switched else 'already'} "
+        f"{g.compute_partition}/{g.memory_partition} "
+        f"({len(g.partitions) or 1} device(s))"
+    )
+    return 0
+
+
+def cmd_health(args) -> int:
+    lib = _lib(args)
+    rc = 0
+    for g in lib.enumerate():
+        h = lib.health_check(g.index)
+        print(f"{g.canonical_name}: {h}")
+        if h.get("status") != "healthy":
+            rc = 1
+    if args.probe:
+        from . import _hiphealth
+
+        for d in range(_hiphealth.device_count()):
+            chk = _hiphealth.mfma_check(d)
+            bw = _hiphealth.bandwidth_gbs(d, 256, 5)
+            print(f"hip:{d}: mfma_ok={chk['ok']} bandwidth={bw:.0f}GB/s")
+            if not chk["ok"]:
+                rc = 1
+    return rc
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser("amd-dra-ctl")
+    ap.add_argument("--hal", default="amdsmi", choices=["amdsmi", "fake"])
+    sub = ap.add_subparsers(dest="cmd", required=True)
+    sub.add_parser("list")
+    sub.add_parser("topology")
+    sub.add_parser("slice")
+    pp = sub.add_parser("partition")
+    pp.add_argument("gpu", type=int)
+    pp.add_argument("compute")
+    pp.add_argument("memory", nargs="?", default="NPS1")
+    hp = sub.add_parser("health")
+    hp.add_argument("--probe", action="store_true", help="also run HIP kernels")
+    args = ap.parse_args(argv)
+    return {
+        "list": cmd_list,
+        "topology": cmd_topology,
+        "slice": cmd_slice,
+        "partition": cmd_partition,
+        "health": cmd_health,
+    }[args.cmd](args)
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -34,6 +34,8 @@ setup(
         "console_scripts": [
             "amd-dra-kubeletplugin=k8s_dra_driver_amd.plugin.main:main",
             "amd-dra-controller=k8s_dra_driver_amd.controller.main:main",
+            "amd-dra-ctl=k8s_dra_driver_amd.ctl:main",
+            "amd-dra-workload=k8s_dra_driver_amd.workload:main",
         ]
     },
 )

@@ -1,0 +1,42 @@
+"""amd-dra-ctl CLI tests on the fake HAL."""
+
+import json
+
+from k8s_dra_driver_amd.ctl import main
+
+
+def test_list(capsys):
+    assert main(["--hal", "fake", "list"]) == 0
+    out = capsys.readouterr().out
+    assert "gpu-0: AMD Instinct MI355X [gfx950]" in out
+    assert "mode=SPX/NPS1" in out
+
+
+def test_topology_matrix(capsys):
+    assert main(["--hal", "fake", "topology"]) == 0
+    out = capsys.readouterr().out
+    assert "hives:" in out
+    assert out.count("x") >= 8  # diagonal
+
+
+def test_slice_json(capsys):
+    assert main(["--hal", "fake", "slice"]) == 0
+    devices = json.loads(capsys.readouterr().out)
+    assert len(devices) == 8
+    assert devices[0]["basic"]["attributes"]["gpu.amd.com/type"]["string"] == "gpu"
+
+
+def test_partition_and_list(capsys):
+    assert main(["--hal", "fake", "partition", "0", "cpx"]) == 0
+    out = capsys.readouterr().out
+    assert "switched to CPX/NPS1 (8 device(s))" in out
+
+
+def test_partition_invalid(capsys):
+    assert main(["--hal", "fake", "partition", "0", "SPX", "NPS4"]) == 1
+    assert "refused" in capsys.readouterr().err
+
+
+def test_health(capsys):
+    assert main(["--hal", "fake", "health"]) == 0
+    assert "healthy" in capsys.readouterr().out

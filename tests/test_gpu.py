@@ -157,6 +157,47 @@ class TestHipHealth:
         assert tf > 500, f"MFMA throughput {tf} TF/s too low"
 
 
+class TestCuMaskEnforcement:
+    """HSA_CU_MASK is the SharedCompute supervisor's spatial-isolation
+    mechanism (docs/sharing.md); verify it really constrains compute on
+    gfx950: 64 of 256 CUs must cut MFMA throughput to ~1/4."""
+
+    def _tflops_with_mask(self, mask_env):
+        import json
+        import subprocess
+        import sys
+
+        code = (
+            "import sys, json;"
+            f"sys.path.insert(0, {os.getcwd()!r});"
+            "import torch;"
+            "from k8s_dra_driver_amd import _hiphealth;"
+            "print(json.dumps(_hiphealth.mfma_tflops(0, 1024, 2048)))"
+        )
+        env = dict(os.environ)
+        if mask_env:
+            env["HSA_CU_MASK"] = mask_env
+        out = subprocess.run(
+            [sys.executable, "-c", code],
+            capture_output=True,
+            text=True,
+            env=env,
+            timeout=300,
+        )
+        assert out.returncode == 0, out.stderr[-2000:]
+        return json.loads(out.stdout.strip().splitlines()[-1])
+
+    def test_cu_mask_quarters_throughput(self):
+        from k8s_dra_driver_amd.sharing.shared import cu_mask_hex
+
+        full = self._tflops_with_mask(None)
+        quarter_mask = f"0:{cu_mask_hex(0, 64, 256)}"
+        quarter = self._tflops_with_mask(quarter_mask)
+        ratio = quarter / full
+        # expect ~0.25; allow clocks/binding slack
+        assert 0.15 < ratio < 0.45, (full, quarter, ratio)
+
+
 class TestTorchInterop:
     def test_torch_sees_gpu(self):
         import torch

@@ -53,8 +53,11 @@ def _rocm_version() -> str:
 
 
 class AmdSmiDeviceLib(DeviceLib):
-    def __init__(self, sysfs_root: str = "/sys"):
+    def __init__(self, sysfs_root: str = "/sys", *, ext=None):
+        """``ext``: test seam — an object with the _amdhal surface; the
+        real native extension is imported when omitted."""
         self._ext = None
+        self._ext_override = ext
         self.topology = KfdTopology(sysfs_root)
         self._timeslice: Dict[int, Optional[int]] = {}
         self._rocm = _rocm_version()
@@ -62,6 +65,10 @@ class AmdSmiDeviceLib(DeviceLib):
     # -- lifecycle ---------------------------------------------------------
     def open(self) -> None:
         if self._ext is not None:
+            return
+        if self._ext_override is not None:
+            self._ext_override.init()
+            self._ext = self._ext_override
             return
         try:
             from .. import _amdhal  # in-tree native extension

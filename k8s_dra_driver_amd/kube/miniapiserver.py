@@ -1,0 +1,175 @@
+"""Mini apiserver: a tiny HTTP server speaking just enough of the
+Kubernetes REST API for this driver's binaries to run without a cluster.
+
+Powers the multi-process e2e tests and clusterless demos: start it, point
+``amd-dra-kubeletplugin --kubeconfig <generated>`` and
+``amd-dra-controller`` at it, and drive claims end-to-end. State is the
+in-memory :class:`k8s_dra_driver_amd.kube.client.InMemoryKube` store, so
+assertions can inspect it directly in-process.
+"""
+
+from __future__ import annotations
+
+import json
+import re
+import threading
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from typing import Optional
+
+from .client import Conflict, InMemoryKube, NotFound
+
+_CLAIM_RE = re.compile(
+    r"^/apis/resource\.k8s\.io/v1beta1/namespaces/([^/]+)/resourceclaims/([^/]+?)(/status)?$"
+)
+_SLICE_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta1/resourceslices(?:/([^/]+))?$")
+_CLAIMS_ALL = "/apis/resource.k8s.io/v1beta1/resourceclaims"
+_CLASSES = "/apis/resource.k8s.io/v1beta1/deviceclasses"
+_NODE_RE = re.compile(r"^/api/v1/nodes/([^/]+)$")
+
+
+class MiniApiServer:
+    def __init__(self, store: Optional[InMemoryKube] = None, port: int = 0):
+        self.store = store or InMemoryKube()
+        outer = self
+
+        class Handler(BaseHTTPRequestHandler):
+            def log_message(self, *a):
+                pass
+
+            def _json(self, code, obj):
+                data = json.dumps(obj).encode()
+                self.send_response(code)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Content-Length", str(len(data)))
+                self.end_headers()
+                self.wfile.write(data)
+
+            def _body(self):
+                n = int(self.headers.get("Content-Length") or 0)
+                return json.loads(self.rfile.read(n)) if n else None
+
+            def do_GET(self):
+                path = self.path.split("?")[0]
+                m = _CLAIM_RE.match(path)
+                if m:
+                    try:
+                        return self._json(
+                            200, outer.store.get_resource_claim(m.group(1), m.group(2))
+                        )
+                    except NotFound:
+                        return self._json(404, {"reason": "NotFound"})
+                if path == _CLAIMS_ALL:
+                    return self._json(
+                        200, {"items": outer.store.list_resource_claims()}
+                    )
+                if path == _CLASSES:
+                    return self._json(
+                        200, {"items": outer.store.get_device_classes()}
+                    )
+                m = _SLICE_RE.match(path)
+                if m and not m.group(1):
+                    return self._json(
+                        200, {"items": outer.store.list_resource_slices()}
+                    )
+                m = _NODE_RE.match(path)
+                if m:
+                    try:
+                        return self._json(200, outer.store.get_node(m.group(1)))
+                    except NotFound:
+                        return self._json(404, {})
+                self._json(404, {"path": path})
+
+            def do_POST(self):
+                path = self.path.split("?")[0]
+                if _SLICE_RE.match(path):
+                    try:
+                        return self._json(
+                            201, outer.store.create_resource_slice(self._body())
+                        )
+                    except Conflict:
+                        return self._json(409, {})
+                self._json(404, {})
+
+            def do_PUT(self):
+                path = self.path.split("?")[0]
+                m = _CLAIM_RE.match(path)
+                if m and m.group(3):  # /status
+                    try:
+                        return self._json(
+                            200,
+                            outer.store.update_resource_claim_status(self._body()),
+                        )
+                    except NotFound:
+                        return self._json(404, {})
+                m = _SLICE_RE.match(path)
+                if m and m.group(1):
+                    try:
+                        return self._json(
+                            200, outer.store.update_resource_slice(self._body())
+                        )
+                    except NotFound:
+                        return self._json(404, {})
+                    except Conflict:
+                        return self._json(409, {})
+                self._json(404, {})
+
+            def do_DELETE(self):
+                path = self.path.split("?")[0]
+                m = _SLICE_RE.match(path)
+                if m and m.group(1):
+                    outer.store.delete_resource_slice(m.group(1))
+                    return self._json(200, {})
+                self._json(404, {})
+
+            def do_PATCH(self):
+                path = self.path.split("?")[0]
+                m = _NODE_RE.match(path)
+                if m:
+                    patch = self._body()
+                    try:
+                        return self._json(
+                            200,
+                            outer.store.patch_node_labels(
+                                m.group(1),
+                                (patch.get("metadata") or {}).get("labels") or {},
+                            ),
+                        )
+                    except NotFound:
+                        return self._json(404, {})
+                self._json(404, {})
+
+        self._server = ThreadingHTTPServer(("127.0.0.1", port), Handler)
+        self._thread: Optional[threading.Thread] = None
+
+    @property
+    def url(self) -> str:
+        return f"http://127.0.0.1:{self._server.server_address[1]}"
+
+    def write_kubeconfig(self, path: str) -> str:
+        with open(path, "w") as f:
+            json.dump(
+                {
+                    "current-context": "mini",
+                    "contexts": [
+                        {"name": "mini", "context": {"cluster": "mini", "user": "u"}}
+                    ],
+                    "clusters": [
+                        {"name": "mini", "cluster": {"server": self.url}}
+                    ],
+                    "users": [{"name": "u", "user": {"token": "dev"}}],
+                },
+                f,
+            )
+        return path
+
+    def start(self) -> "MiniApiServer":
+        self._thread = threading.Thread(
+            target=self._server.serve_forever, name="mini-apiserver", daemon=True
+        )
+        self._thread.start()
+        return self
+
+    def stop(self) -> None:
+        self._server.shutdown()
+        if self._thread:
+            self._thread.join(timeout=5)

@@ -1,0 +1,76 @@
+"""Guards the bench.py driver contract: flags, JSON schema, single line."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.timeout(300)
+def test_bench_json_contract():
+    out = subprocess.run(
+        [
+            sys.executable,
+            os.path.join(REPO, "bench.py"),
+            "--steps",
+            "2",
+            "--warmup",
+            "1",
+            "--pods-per-step",
+            "4",
+        ],
+        capture_output=True,
+        text=True,
+        cwd=REPO,
+        timeout=280,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    for key in (
+        "metric",
+        "value",
+        "unit",
+        "n_gpus",
+        "steps",
+        "warmup",
+        "ms_per_step",
+        "higher_is_better",
+        "scaling",
+        "vs_baseline",
+        "dtype",
+        "data",
+        "config",
+    ):
+        assert key in d, key
+    assert d["metric"] == "gpu_pods_scheduled_per_sec"
+    assert d["n_gpus"] == 1
+    assert d["steps"] == 2 and d["warmup"] == 1
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["vs_baseline"] is None  # reference publishes no number
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    cfg = d["config"]
+    assert cfg["model"] == "dra-claim-lifecycle"
+    assert cfg["hal"] in ("fake", "amdsmi")
+    assert cfg["alloc_prepare_p50_ms"] > 0
+
+
+def test_metrics_histograms():
+    from k8s_dra_driver_amd.metrics.prom import PluginMetrics
+    from prometheus_client import generate_latest
+
+    m = PluginMetrics()
+    with m.time_prepare():
+        pass
+    m.prepared_claims.inc()
+    m.allocatable_devices.set(8)
+    text = generate_latest(m.registry).decode()
+    assert "dra_prepare_seconds_bucket" in text
+    assert "dra_prepared_claims_total 1.0" in text
+    assert "dra_allocatable_devices 8.0" in text

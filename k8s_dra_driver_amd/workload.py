@@ -27,11 +27,74 @@ def visible_devices() -> dict:
     }
 
 
+def _allreduce_worker(rank, world, nelem, results):
+    """One process per visible GPU; RCCL ring all-reduce over xGMI."""
+    import os as _os
+    import time
+
+    import torch
+    import torch.distributed as dist
+
+    _os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    _os.environ.setdefault("MASTER_PORT", "29781")
+    dist.init_process_group("nccl", rank=rank, world_size=world)
+    torch.cuda.set_device(rank)
+    x = torch.ones(nelem, dtype=torch.bfloat16, device=f"cuda:{rank}")
+    for _ in range(3):  # warmup
+        dist.all_reduce(x)
+    torch.cuda.synchronize()
+    iters = 20
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        dist.all_reduce(x)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    # ring all-reduce moves 2*(n-1)/n of the buffer per GPU per iteration
+    bytes_moved = 2 * (world - 1) / world * nelem * 2 * iters
+    ok = bool(torch.all(x == float(world) ** (iters + 3)).item()) if world > 1 else True
+    results[rank] = (bytes_moved / dt / 1e9, ok)
+    dist.destroy_process_group()
+
+
+def run_allreduce(size_mb: int) -> int:
+    """RCCL all-reduce across every visible GPU (the xGMI fabric demo for
+    multi-GPU topology claims, gpu-test7)."""
+    import torch
+    import torch.multiprocessing as mp
+
+    world = torch.cuda.device_count()
+    if world == 0:
+        print("ERROR: no HIP devices", file=sys.stderr)
+        return 1
+    nelem = size_mb * 1024 * 1024 // 2  # bf16
+    if world == 1:
+        print("1 visible GPU: all-reduce is a no-op; device healthy")
+        return 0
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(
+            _allreduce_worker,
+            args=(world, nelem, results),
+            nprocs=world,
+            join=True,
+        )
+        per_gpu = dict(results)
+    for r, (gbps, ok) in sorted(per_gpu.items()):
+        print(f"rank {r}: busbw {gbps:.1f} GB/s numerics_ok={ok}")
+    return 0 if all(ok for _, ok in per_gpu.values()) else 1
+
+
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser("amd-dra-workload")
     ap.add_argument("--list", action="store_true", help="print visible devices (nvidia-smi -L analog)")
     ap.add_argument("--benchmark", action="store_true", help="run bandwidth + MFMA probes")
     ap.add_argument("--burn-ms", type=int, default=0, help="occupancy burn duration")
+    ap.add_argument(
+        "--allreduce-mb",
+        type=int,
+        default=0,
+        help="RCCL all-reduce of this buffer size across visible GPUs",
+    )
     args = ap.parse_args(argv)
 
     info = visible_devices()
@@ -52,6 +115,8 @@ def main(argv=None) -> int:
             if args.burn_ms:
                 ms = _hiphealth.burn_ms(d, args.burn_ms)
                 print(f"device {d}: burned {ms:.0f} ms")
+    if args.allreduce_mb:
+        return run_allreduce(args.allreduce_mb)
     return 0
 
 

@@ -29,7 +29,7 @@ runtime error means *no match* (the allocator treats it so).
 from __future__ import annotations
 
 import re
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 from ..api.types import parse_quantity_bytes
 

@@ -27,7 +27,7 @@ import re
 from dataclasses import dataclass, field
 from typing import Any, Dict, List, Optional
 
-from ..partition.catalog import COMPUTE_MODES, MEMORY_MODES, validate_mode_combo
+from ..partition.catalog import validate_mode_combo
 
 API_GROUP = "resource.gpu.amd.com"
 API_VERSION = "v1alpha1"

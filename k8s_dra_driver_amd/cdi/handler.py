@@ -17,7 +17,7 @@ import os
 import re
 from typing import Dict, Iterable, List, Optional
 
-from ..hal.model import AllocatableDevice, GpuInfo
+from ..hal.model import AllocatableDevice
 from .spec import (
     CDIDevice,
     CDISpec,

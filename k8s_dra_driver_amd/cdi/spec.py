@@ -25,7 +25,7 @@ import json
 import os
 import tempfile
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import List
 
 # CDI versions: 0.5.0 baseline; 0.6.0 adds annotations/class-level deviceNodes
 # semantics we use. Stamp the minimum the spec actually needs (reference

@@ -21,11 +21,7 @@ from __future__ import annotations
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
-from ..partition.catalog import (
-    GFX950_ARCH,
-    GFX950_XGMI_LINKS_PER_GPU,
-    PartitionProfile,
-)
+from ..partition.catalog import GFX950_ARCH, PartitionProfile
 
 #: attribute/capacity qualified-name domain (reference uses gpu.nvidia.com)
 DOMAIN = "gpu.amd.com"

@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import os
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 KFD_IOLINK_TYPE_PCIE = 2
 KFD_IOLINK_TYPE_XGMI = 11

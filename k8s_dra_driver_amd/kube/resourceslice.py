@@ -18,7 +18,7 @@ from __future__ import annotations
 import hashlib
 import json
 import threading
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from .client import Conflict, KubeClient, NotFound
 

@@ -23,7 +23,7 @@ in :mod:`k8s_dra_driver_amd.partition.manager`.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Sequence, Tuple
 
 # Chip constants for AMD Instinct MI355X (gfx950).

@@ -27,7 +27,6 @@ from ..hal.base import DeviceLib
 from ..kube.client import KubeClient, NotFound
 from ..kube.resourceslice import ResourceSlicePublisher
 from ..metrics.prom import PluginMetrics
-from ..partition.manager import PartitionManager
 from ..sharing.shared import SharedComputeManager
 from ..sharing.timeslice import TimeSlicingManager
 from ..state.checkpoint import CheckpointStore

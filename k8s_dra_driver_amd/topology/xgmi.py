@@ -14,7 +14,7 @@ over one representation.
 from __future__ import annotations
 
 import itertools
-from typing import Dict, List, Sequence, Tuple
+from typing import Dict, List, Sequence
 
 from ..hal.model import DOMAIN
 

@@ -146,7 +146,30 @@ class DeviceState:
     # ------------------------------------------------------------------
     def _recover(self) -> None:
         recovered = self.checkpoints.list_all()
+        #: claims whose checkpointed devices no longer exist on the node
+        #: (hardware/partition drift while the plugin was down) — the
+        #: stale-resume gap the reference leaves unhandled (SURVEY §5.4:
+        #: checkpoints are never re-validated against live hardware).
+        #: They are kept (their pods may still run) but flagged for
+        #: operators and the health/metrics surface.
+        self.stale_claims: Dict[str, List[str]] = {}
+        with self._registry_lock:
+            live_names = set(self._allocatable)
         for uid, pc in recovered.items():
+            missing = [
+                d.device_name
+                for d in pc.devices
+                if d.device_name not in live_names
+            ]
+            if missing:
+                self.stale_claims[uid] = missing
+                log.warning(
+                    "recovered claim %s references device(s) %s that no "
+                    "longer exist (partition/hardware drift while the "
+                    "plugin was down); claim kept, flagged stale",
+                    uid,
+                    missing,
+                )
             for dev in pc.devices:
                 if dev.parent_gpu_index >= 0:
                     self._gpu_holders.setdefault(dev.parent_gpu_index, set()).add(uid)

@@ -40,15 +40,24 @@ class AllocationError(RuntimeError):
 class DeviceClass:
     name: str
     selectors: List[str] = field(default_factory=list)
+    #: opaque config entries a DeviceClass attaches to every allocation
+    #: using it (merged as source=FromClass, lower precedence than claim
+    #: config — reference device_state.go:462-477 consumes both)
+    config: List[dict] = field(default_factory=list)
 
     @classmethod
     def from_obj(cls, obj: dict) -> "DeviceClass":
+        spec = obj.get("spec", {})
         sels = [
             s["cel"]["expression"]
-            for s in (obj.get("spec", {}).get("selectors") or [])
+            for s in (spec.get("selectors") or [])
             if "cel" in s
         ]
-        return cls(name=obj["metadata"]["name"], selectors=sels)
+        return cls(
+            name=obj["metadata"]["name"],
+            selectors=sels,
+            config=list(spec.get("config") or []),
+        )
 
 
 #: The driver's default DeviceClasses (Helm chart parity:
@@ -89,14 +98,18 @@ class AllocationResult:
     device: str
     pool: str
     driver: str = DRIVER_NAME
+    admin_access: bool = False
 
     def to_obj(self) -> dict:
-        return {
+        out = {
             "request": self.request,
             "driver": self.driver,
             "pool": self.pool,
             "device": self.device,
         }
+        if self.admin_access:
+            out["adminAccess"] = True
+        return out
 
 
 class Allocator:
@@ -157,7 +170,10 @@ class Allocator:
         avail = [d for d in devices if d["name"] not in in_use]
         per_request: List[Tuple[dict, List[dict], int]] = []
         for r in requests:
-            cands = self.candidates_for_request(r, avail)
+            # adminAccess requests see every device, in-use included
+            # (monitoring claims don't consume exclusivity)
+            pool_devices = devices if r.get("adminAccess") else avail
+            cands = self.candidates_for_request(r, pool_devices)
             mode = r.get("allocationMode", "ExactCount")
             count = len(cands) if mode == "All" else int(r.get("count", 1))
             if mode != "All" and len(cands) < count:
@@ -178,7 +194,10 @@ class Allocator:
             for d in devs:
                 out.append(
                     AllocationResult(
-                        request=r.get("name", ""), device=d["name"], pool=pool
+                        request=r.get("name", ""),
+                        device=d["name"],
+                        pool=pool,
+                        admin_access=bool(r.get("adminAccess")),
                     )
                 )
         return out
@@ -302,7 +321,27 @@ class Allocator:
         results = self.allocate(
             claim.get("spec", {}), devices, pool=pool, in_use=in_use
         )
+        # DeviceClass-attached config merges in first with source=FromClass
+        # (lower precedence than claim config — the kube-scheduler behavior
+        # the reference consumes, device_state.go:462-477).
         config = []
+        requests = claim.get("spec", {}).get("devices", {}).get("requests") or []
+        seen_classes = []
+        for r in requests:
+            cls = self.classes.get(r.get("deviceClassName", ""))
+            if cls is None or not cls.config or cls.name in seen_classes:
+                continue
+            seen_classes.append(cls.name)
+            req_names = [
+                rq.get("name", "")
+                for rq in requests
+                if rq.get("deviceClassName") == cls.name
+            ]
+            for c in cls.config:
+                entry = dict(c)
+                entry["source"] = "FromClass"
+                entry.setdefault("requests", req_names)
+                config.append(entry)
         for c in claim.get("spec", {}).get("devices", {}).get("config") or []:
             entry = dict(c)
             entry.setdefault("source", "FromClaim")

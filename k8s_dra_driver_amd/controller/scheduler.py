@@ -75,6 +75,8 @@ class ClaimScheduler:
             for r in (alloc.get("devices") or {}).get("results") or []:
                 if r.get("driver") != self.driver:
                     continue
+                if r.get("adminAccess"):
+                    continue  # monitoring claims don't consume devices
                 out.setdefault(r.get("pool", ""), set()).add(r.get("device"))
         return out
 

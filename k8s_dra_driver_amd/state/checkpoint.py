@@ -41,6 +41,8 @@ class PreparedDevice:
     parent_gpu_index: int = -1
     kind: str = "gpu"  # gpu | partition
     device_uuid: str = ""
+    #: adminAccess results don't hold the GPU (no drain/exclusivity)
+    admin: bool = False
 
 
 @dataclass

@@ -228,7 +228,9 @@ class DeviceState:
                 fut.result()  # join the overlapped claim-spec write
             with self._registry_lock:
                 for dev in prepared.devices:
-                    if dev.parent_gpu_index >= 0:
+                    # adminAccess (monitoring) claims neither block
+                    # repartition drains nor count as exclusive holders
+                    if dev.parent_gpu_index >= 0 and not dev.admin:
                         self._gpu_holders.setdefault(
                             dev.parent_gpu_index, set()
                         ).add(info.uid)
@@ -427,6 +429,7 @@ class DeviceState:
                     parent_gpu_index=dev.parent_gpu.index,
                     kind=dev.kind,
                     device_uuid=dev.uuid,
+                    admin=bool(r.get("adminAccess")),
                 )
             )
         return prepared

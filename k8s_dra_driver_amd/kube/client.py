@@ -42,6 +42,9 @@ class KubeClient(abc.ABC):
         """DeviceClass objects (controller allocator input); optional."""
         return []
 
+    def create_event(self, namespace: str, event: dict) -> None:
+        """Emit a Kubernetes Event (operator visibility); best-effort."""
+
     @abc.abstractmethod
     def create_resource_slice(self, obj: dict) -> dict: ...
 
@@ -72,6 +75,7 @@ class InMemoryKube(KubeClient):
         self.resource_slices: Dict[str, dict] = {}
         self.nodes: Dict[str, dict] = {}
         self.device_classes: Dict[str, dict] = {}
+        self.events: List[dict] = []
         self._rv = 0
         #: watch hooks: fn(kind, verb, obj)
         self.watchers: List[Callable[[str, str, dict], None]] = []
@@ -130,6 +134,10 @@ class InMemoryKube(KubeClient):
         with self._lock:
             self.device_classes[obj["metadata"]["name"]] = copy.deepcopy(obj)
             return obj
+
+    def create_event(self, namespace: str, event: dict) -> None:
+        with self._lock:
+            self.events.append({"namespace": namespace, **copy.deepcopy(event)})
 
     def create_resource_slice(self, obj: dict) -> dict:
         with self._lock:

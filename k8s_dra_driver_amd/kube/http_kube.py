@@ -162,6 +162,12 @@ class HttpKube(KubeClient):
         out = self._req("GET", f"{RESOURCE_V1BETA1}/deviceclasses")
         return out.get("items", [])
 
+    def create_event(self, namespace: str, event: dict) -> None:
+        try:
+            self._req("POST", f"{CORE_V1}/namespaces/{namespace}/events", event)
+        except Exception:
+            pass  # events are best-effort
+
     def get_node(self, name: str) -> dict:
         return self._req("GET", f"{CORE_V1}/nodes/{name}")
 

@@ -167,6 +167,7 @@ class Driver:
                 )
             except PrepareError as e:
                 self.metrics.prepare_errors.inc()
+                self._emit_failure_event(ref, str(e))
                 return ClaimResult([], str(e))
             except Exception as e:
                 log.exception("prepare %s failed", ref.uid)
@@ -174,6 +175,33 @@ class Driver:
                 return ClaimResult([], f"internal error preparing claim: {e}")
             self.metrics.prepared_claims.inc()
             return ClaimResult(devices)
+
+    def _emit_failure_event(self, ref: ClaimRef, message: str) -> None:
+        """Kubernetes Event on prepare failure (operator visibility the
+        reference lacks — failures only surface in kubelet logs there)."""
+        try:
+            self.kube.create_event(
+                ref.namespace,
+                {
+                    "metadata": {
+                        "generateName": "amd-dra-prepare-",
+                        "namespace": ref.namespace,
+                    },
+                    "type": "Warning",
+                    "reason": "PrepareFailed",
+                    "message": message[:1024],
+                    "involvedObject": {
+                        "apiVersion": "resource.k8s.io/v1beta1",
+                        "kind": "ResourceClaim",
+                        "namespace": ref.namespace,
+                        "name": ref.name,
+                        "uid": ref.uid,
+                    },
+                    "source": {"component": DRIVER_NAME},
+                },
+            )
+        except Exception:
+            log.debug("event emission failed", exc_info=True)
 
     # ------------------------------------------------------------------
     def cleanup_orphans(self) -> List[str]:

@@ -82,3 +82,19 @@ def test_specless_cdi_file_cleaned(tmp_path):
     cleaned = driver.cleanup_orphans()
     assert "ghost-uid" in cleaned
     assert driver.state.cdi.list_claim_spec_uids() == []
+
+
+def test_prepare_failure_emits_event(tmp_path):
+    kube, driver = make_stack(tmp_path)
+    put_claim(kube, "uid-ev")
+    kube.resource_claims["d/c1"]["status"]["allocation"]["devices"]["results"][0][
+        "device"
+    ] = "gpu-404"
+    kube.put_resource_claim(kube.resource_claims["d/c1"])
+    res = driver.node_prepare_resources([ClaimRef("d", "c1", "uid-ev")])["uid-ev"]
+    assert "not found" in res.error
+    assert kube.events, "no Event emitted"
+    ev = kube.events[-1]
+    assert ev["reason"] == "PrepareFailed"
+    assert ev["involvedObject"]["name"] == "c1"
+    assert ev["type"] == "Warning"

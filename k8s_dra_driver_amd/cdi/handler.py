@@ -52,12 +52,19 @@ class CDIHandler:
         *,
         dev_root: str = "",
         driver_version: str = "",
+        rocm_mount: str = "",
     ):
         self.cdi_root = cdi_root
         # dev_root: prefix of host /dev (containerized-driver-root analog,
         # reference root.go:76-98); "" = host /dev directly.
         self.dev_root = dev_root.rstrip("/")
         self.driver_version = driver_version
+        # rocm_mount: host ROCm path to bind into containers read-only —
+        # the driver-library-injection analog (reference cdi.go:158-227)
+        # for workload images that ship no ROCm userspace. "" = off
+        # (images normally carry their own ROCm; the amdgpu/KFD ABI is
+        # stable across versions).
+        self.rocm_mount = rocm_mount.rstrip("/")
 
     # -- paths -------------------------------------------------------------
     def _base_spec_path(self) -> str:
@@ -116,6 +123,13 @@ class CDIHandler:
         )
         if self.driver_version:
             common.env.append(f"AMD_DRIVER_VERSION={self.driver_version}")
+        if self.rocm_mount:
+            from .spec import Mount
+
+            common.mounts.append(
+                Mount(host_path=self.rocm_mount, container_path="/opt/rocm")
+            )
+            common.env.append("ROCM_PATH=/opt/rocm")
         spec = CDISpec(kind=DEVICE_KIND, common_edits=common)
         for dev in devices:
             spec.devices.append(

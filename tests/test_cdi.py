@@ -132,3 +132,15 @@ def test_dev_root_prefix(tmp_path, fake_lib):
     kfd = spec["containerEdits"]["deviceNodes"][0]
     assert kfd["hostPath"] == "/driver-root/dev/kfd"
     assert kfd["path"] == "/dev/kfd"
+
+
+def test_rocm_mount_option(tmp_path, fake_lib):
+    h = CDIHandler(cdi_root=str(tmp_path), rocm_mount="/opt/rocm-7.2.0")
+    path = h.create_standard_spec(_allocatable(fake_lib))
+    spec = read_spec_file(path)
+    common = spec["containerEdits"]
+    mounts = common["mounts"]
+    assert mounts[0]["hostPath"] == "/opt/rocm-7.2.0"
+    assert mounts[0]["containerPath"] == "/opt/rocm"
+    assert "ROCM_PATH=/opt/rocm" in common["env"]
+    assert spec["cdiVersion"] == "0.6.0"  # mounts bump the min version

@@ -46,7 +46,43 @@ __global__ void copy_f4(const float4* __restrict__ src,
   for (; i < n; i += stride) dst[i] = src[i];
 }
 
-double bandwidth_gbs(int device, int mib, int iters) {
+// Non-temporal variant: streamed data is read/written once — bypassing
+// cache retention buys back bandwidth on pure streams (the guide's
+// nt-weights pattern applied to a copy).
+__global__ void copy_f4_nt(const float4* __restrict__ src,
+                           float4* __restrict__ dst, size_t n) {
+  size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float4 v = __builtin_nontemporal_load(&src[i]);
+    __builtin_nontemporal_store(v, &dst[i]);
+  }
+}
+
+static double time_copy(void (*kernel)(const float4*, float4*, size_t),
+                        const float4* src, float4* dst, size_t n,
+                        unsigned blocks, int iters, size_t bytes) {
+  dim3 block(256), grid(blocks);
+  hipLaunchKernelGGL(kernel, grid, block, 0, 0, src, dst, n);  // warmup
+  HIP_CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL(kernel, grid, block, 0, 0, src, dst, n);
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  double gb = 2.0 * (double)bytes * iters / 1e9;  // read + write
+  return gb / (ms / 1e3);
+}
+
+double bandwidth_gbs(int device, int mib, int iters, unsigned blocks,
+                     bool nt) {
   HIP_CHECK(hipSetDevice(device));
   size_t bytes = (size_t)mib << 20;
   size_t n = bytes / sizeof(float4);
@@ -54,28 +90,34 @@ double bandwidth_gbs(int device, int mib, int iters) {
   HIP_CHECK(hipMalloc(&src, bytes));
   HIP_CHECK(hipMalloc(&dst, bytes));
   HIP_CHECK(hipMemset(src, 1, bytes));
-  dim3 block(256);
-  // >> 256 workgroups to fill 8 XCDs; cap so the tail loop stays short.
-  dim3 grid((unsigned)std::min<size_t>((n + 255) / 256, 8192));
-  // warmup
-  hipLaunchKernelGGL(copy_f4, grid, block, 0, 0, src, dst, n);
-  HIP_CHECK(hipDeviceSynchronize());
-  hipEvent_t t0, t1;
-  HIP_CHECK(hipEventCreate(&t0));
-  HIP_CHECK(hipEventCreate(&t1));
-  HIP_CHECK(hipEventRecord(t0));
-  for (int i = 0; i < iters; ++i)
-    hipLaunchKernelGGL(copy_f4, grid, block, 0, 0, src, dst, n);
-  HIP_CHECK(hipEventRecord(t1));
-  HIP_CHECK(hipEventSynchronize(t1));
-  float ms = 0;
-  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
-  HIP_CHECK(hipEventDestroy(t0));
-  HIP_CHECK(hipEventDestroy(t1));
+  if (blocks == 0)
+    blocks = (unsigned)std::min<size_t>((n + 255) / 256, 8192);
+  double gbs = time_copy(nt ? copy_f4_nt : copy_f4, src, dst, n, blocks,
+                         iters, bytes);
   HIP_CHECK(hipFree(src));
   HIP_CHECK(hipFree(dst));
-  double gb = 2.0 * (double)bytes * iters / 1e9;  // read + write
-  return gb / (ms / 1e3);
+  return gbs;
+}
+
+py::dict bandwidth_sweep(int device, int mib, int iters) {
+  // variant sweep used to pick the probe's defaults on real hardware
+  HIP_CHECK(hipSetDevice(device));
+  size_t bytes = (size_t)mib << 20;
+  size_t n = bytes / sizeof(float4);
+  float4 *src = nullptr, *dst = nullptr;
+  HIP_CHECK(hipMalloc(&src, bytes));
+  HIP_CHECK(hipMalloc(&dst, bytes));
+  HIP_CHECK(hipMemset(src, 1, bytes));
+  py::dict out;
+  for (unsigned blocks : {1024u, 2048u, 4096u, 8192u, 16384u}) {
+    out[py::str("plain_" + std::to_string(blocks))] =
+        time_copy(copy_f4, src, dst, n, blocks, iters, bytes);
+    out[py::str("nt_" + std::to_string(blocks))] =
+        time_copy(copy_f4_nt, src, dst, n, blocks, iters, bytes);
+  }
+  HIP_CHECK(hipFree(src));
+  HIP_CHECK(hipFree(dst));
+  return out;
 }
 
 // ---------------------------------------------------------------------------
@@ -236,7 +278,10 @@ PYBIND11_MODULE(_hiphealth, m) {
   m.def("device_count", &device_count);
   m.def("device_info", &device_info, py::arg("device") = 0);
   m.def("bandwidth_gbs", &bandwidth_gbs, py::arg("device") = 0,
-        py::arg("mib") = 1024, py::arg("iters") = 10);
+        py::arg("mib") = 1024, py::arg("iters") = 10,
+        py::arg("blocks") = 0, py::arg("nt") = false);
+  m.def("bandwidth_sweep", &bandwidth_sweep, py::arg("device") = 0,
+        py::arg("mib") = 1024, py::arg("iters") = 5);
   m.def("mfma_check", &mfma_check, py::arg("device") = 0);
   m.def("mfma_tflops", &mfma_tflops, py::arg("device") = 0,
         py::arg("iters") = 8192, py::arg("blocks") = 2048);

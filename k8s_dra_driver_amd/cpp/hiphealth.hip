@@ -49,13 +49,18 @@ __global__ void copy_f4(const float4* __restrict__ src,
 // Non-temporal variant: streamed data is read/written once — bypassing
 // cache retention buys back bandwidth on pure streams (the guide's
 // nt-weights pattern applied to a copy).
+typedef float vf4 __attribute__((ext_vector_type(4)));  // nt builtins need
+                                                        // a native vector
+
 __global__ void copy_f4_nt(const float4* __restrict__ src,
                            float4* __restrict__ dst, size_t n) {
+  const vf4* __restrict__ s = reinterpret_cast<const vf4*>(src);
+  vf4* __restrict__ d = reinterpret_cast<vf4*>(dst);
   size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
   size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
-    float4 v = __builtin_nontemporal_load(&src[i]);
-    __builtin_nontemporal_store(v, &dst[i]);
+    vf4 v = __builtin_nontemporal_load(&s[i]);
+    __builtin_nontemporal_store(v, &d[i]);
   }
 }
 

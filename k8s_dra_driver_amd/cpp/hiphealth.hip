@@ -64,6 +64,30 @@ __global__ void copy_f4_nt(const float4* __restrict__ src,
   }
 }
 
+// 4-way grid-stride unroll: four independent coalesced accesses in
+// flight per lane (guide: keep >=8 loads/lane outstanding on streams).
+__global__ void copy_f4_nt_u4(const float4* __restrict__ src,
+                              float4* __restrict__ dst, size_t n) {
+  const vf4* __restrict__ s = reinterpret_cast<const vf4*>(src);
+  vf4* __restrict__ d = reinterpret_cast<vf4*>(dst);
+  size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i + 3 * stride < n; i += 4 * stride) {
+    vf4 a = __builtin_nontemporal_load(&s[i]);
+    vf4 b = __builtin_nontemporal_load(&s[i + stride]);
+    vf4 c = __builtin_nontemporal_load(&s[i + 2 * stride]);
+    vf4 e = __builtin_nontemporal_load(&s[i + 3 * stride]);
+    __builtin_nontemporal_store(a, &d[i]);
+    __builtin_nontemporal_store(b, &d[i + stride]);
+    __builtin_nontemporal_store(c, &d[i + 2 * stride]);
+    __builtin_nontemporal_store(e, &d[i + 3 * stride]);
+  }
+  for (; i < n; i += stride) {
+    vf4 v = __builtin_nontemporal_load(&s[i]);
+    __builtin_nontemporal_store(v, &d[i]);
+  }
+}
+
 static double time_copy(void (*kernel)(const float4*, float4*, size_t),
                         const float4* src, float4* dst, size_t n,
                         unsigned blocks, int iters, size_t bytes) {
@@ -95,9 +119,10 @@ double bandwidth_gbs(int device, int mib, int iters, unsigned blocks,
   HIP_CHECK(hipMalloc(&src, bytes));
   HIP_CHECK(hipMalloc(&dst, bytes));
   HIP_CHECK(hipMemset(src, 1, bytes));
-  if (blocks == 0)
-    blocks = (unsigned)std::min<size_t>((n + 255) / 256, 8192);
-  double gbs = time_copy(nt ? copy_f4_nt : copy_f4, src, dst, n, blocks,
+  // defaults from the round-1 hardware sweep (profiles/): nt + 1024 WGs
+  // (4 per CU) beat every larger grid and the cached path.
+  if (blocks == 0) blocks = 1024;
+  double gbs = time_copy(nt ? copy_f4_nt_u4 : copy_f4, src, dst, n, blocks,
                          iters, bytes);
   HIP_CHECK(hipFree(src));
   HIP_CHECK(hipFree(dst));
@@ -114,11 +139,13 @@ py::dict bandwidth_sweep(int device, int mib, int iters) {
   HIP_CHECK(hipMalloc(&dst, bytes));
   HIP_CHECK(hipMemset(src, 1, bytes));
   py::dict out;
-  for (unsigned blocks : {1024u, 2048u, 4096u, 8192u, 16384u}) {
+  for (unsigned blocks : {512u, 1024u, 2048u, 4096u}) {
     out[py::str("plain_" + std::to_string(blocks))] =
         time_copy(copy_f4, src, dst, n, blocks, iters, bytes);
     out[py::str("nt_" + std::to_string(blocks))] =
         time_copy(copy_f4_nt, src, dst, n, blocks, iters, bytes);
+    out[py::str("ntu4_" + std::to_string(blocks))] =
+        time_copy(copy_f4_nt_u4, src, dst, n, blocks, iters, bytes);
   }
   HIP_CHECK(hipFree(src));
   HIP_CHECK(hipFree(dst));
@@ -284,7 +311,7 @@ PYBIND11_MODULE(_hiphealth, m) {
   m.def("device_info", &device_info, py::arg("device") = 0);
   m.def("bandwidth_gbs", &bandwidth_gbs, py::arg("device") = 0,
         py::arg("mib") = 1024, py::arg("iters") = 10,
-        py::arg("blocks") = 0, py::arg("nt") = false);
+        py::arg("blocks") = 0, py::arg("nt") = true);
   m.def("bandwidth_sweep", &bandwidth_sweep, py::arg("device") = 0,
         py::arg("mib") = 1024, py::arg("iters") = 5);
   m.def("mfma_check", &mfma_check, py::arg("device") = 0);

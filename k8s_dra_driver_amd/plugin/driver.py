@@ -61,11 +61,16 @@ class Driver:
         use_tmpfs: Optional[bool] = None,
         max_concurrent_claims: int = 16,
         metrics: Optional[PluginMetrics] = None,
+        device_kinds: Optional[List[str]] = None,
     ):
         self.lib = lib
         self.kube = kube
         self.node_name = node_name
         self.metrics = metrics or PluginMetrics()
+        # which device kinds to publish ("gpu", "partition") — the
+        # reference's --device-classes subsystem gating (driver.go:66,
+        # nvlib.go:113-133); None = all.
+        self.device_kinds = device_kinds
         cdi = CDIHandler(cdi_root=cdi_root)
         checkpoints = CheckpointStore(checkpoint_root)
         shared = SharedComputeManager(
@@ -116,6 +121,7 @@ class Driver:
             d.to_device()
             for d in self.state.allocatable_devices()
             if d.parent_gpu.index not in unhealthy
+            and (self.device_kinds is None or d.kind in self.device_kinds)
         ]
         self.publisher.publish(devices)
         self.metrics.allocatable_devices.set(len(devices))

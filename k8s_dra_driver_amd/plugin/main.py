@@ -48,6 +48,12 @@ def build_parser() -> argparse.ArgumentParser:
         "--cdi-root", default=_env("CDI_ROOT", DEFAULT_CDI_ROOT)
     )
     p.add_argument(
+        "--device-classes",
+        default=_env("DEVICE_CLASSES", "gpu,partition"),
+        help="comma list of device kinds to publish (gpu,partition) — "
+        "subsystem gating parity with the reference",
+    )
+    p.add_argument(
         "--plugin-registration-path",
         default=_env("PLUGIN_REGISTRATION_PATH", DEFAULT_REGISTRY),
     )
@@ -114,6 +120,7 @@ def main(argv=None) -> int:
         cdi_root=args.cdi_root,
         checkpoint_root=os.path.join(args.plugin_path, "state"),
         metrics=metrics,
+        device_kinds=[s.strip() for s in args.device_classes.split(",") if s.strip()],
     )
     driver.startup()
     driver.health.start()  # failure detection -> slice self-healing

@@ -53,3 +53,27 @@ def test_unhealthy_gpu_pulled_from_slices(tmp_path):
     devices = kube.list_resource_slices(DRIVER_NAME)[0]["spec"]["devices"]
     assert len(devices) == 7
     assert "gpu-0" not in [d["name"] for d in devices]
+
+
+def test_device_kinds_gating(tmp_path):
+    """--device-classes parity: publish only the enabled kinds."""
+    lib = FakeDeviceLib()
+    lib.open()
+    lib.set_compute_partition(0, "CPX")
+    kube = InMemoryKube()
+    driver = Driver(
+        lib,
+        kube,
+        node_name="n",
+        cdi_root=str(tmp_path / "cdi"),
+        checkpoint_root=str(tmp_path / "state"),
+        use_tmpfs=False,
+        device_kinds=["partition"],
+    )
+    driver.startup()
+    names = [
+        d["name"]
+        for s in kube.list_resource_slices(DRIVER_NAME)
+        for d in s["spec"]["devices"]
+    ]
+    assert names and all("cpx" in n for n in names)

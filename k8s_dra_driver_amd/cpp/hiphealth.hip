@@ -119,10 +119,12 @@ double bandwidth_gbs(int device, int mib, int iters, unsigned blocks,
   HIP_CHECK(hipMalloc(&src, bytes));
   HIP_CHECK(hipMalloc(&dst, bytes));
   HIP_CHECK(hipMemset(src, 1, bytes));
-  // defaults from the round-1 hardware sweep (profiles/): nt + 1024 WGs
-  // (4 per CU) beat every larger grid and the cached path.
+  // defaults from the round-1 hardware sweeps (profiles/): 1024 WGs
+  // (4 per CU) beats every larger grid; plain and nt tie within run
+  // variance at that size (5495-5590 GB/s) and the 4-way unroll LOSES
+  // ~10% there (measured 4965), so the default is the simple nt loop.
   if (blocks == 0) blocks = 1024;
-  double gbs = time_copy(nt ? copy_f4_nt_u4 : copy_f4, src, dst, n, blocks,
+  double gbs = time_copy(nt ? copy_f4_nt : copy_f4, src, dst, n, blocks,
                          iters, bytes);
   HIP_CHECK(hipFree(src));
   HIP_CHECK(hipFree(dst));

@@ -20,7 +20,7 @@ bench:
 lint:
 	-ruff check k8s_dra_driver_amd/ tests/ bench.py
 
-coverage:
+coverage:  # requires pytest-cov (not in the offline image)
 	$(PYTHON) -m pytest tests/ -q -m "not gpu" --cov=k8s_dra_driver_amd --cov-report=term
 
 image:

@@ -95,6 +95,8 @@ class ControllerManager:
 
             self.scheduler = ClaimScheduler(kube)
         self._stop = threading.Event()
+        self._kick = threading.Event()  # watch-triggered wakeup
+        self._watch = None
         self._thread: Optional[threading.Thread] = None
         self._owned_labels: Dict[str, Dict[str, str]] = {}
 
@@ -157,7 +159,9 @@ class ControllerManager:
 
     # ------------------------------------------------------------------
     def run(self) -> None:
-        """Reconcile loop with transient-error retry (imex.go:132-151)."""
+        """Level-triggered reconcile loop: wakes on watch events (claim
+        churn) or the resync interval; transient-error retry parity with
+        the reference (imex.go:132-151)."""
         backoff = self.poll_interval
         while not self._stop.is_set():
             try:
@@ -166,9 +170,23 @@ class ControllerManager:
             except Exception:
                 log.exception("reconcile failed; retrying in %.0fs", RETRY_SECONDS)
                 backoff = RETRY_SECONDS
-            self._stop.wait(backoff)
+            self._kick.wait(backoff)
+            self._kick.clear()
 
     def start(self) -> None:
+        if self.scheduler is not None:
+            # informer-style wakeup: allocate new claims promptly instead
+            # of waiting out the poll interval (client-go informer analog)
+            def on_claim(event_type: str, obj: dict) -> None:
+                if event_type in ("ADDED", "MODIFIED") and not (
+                    obj.get("status") or {}
+                ).get("allocation"):
+                    self._kick.set()
+
+            try:
+                self._watch = self.kube.watch_resource_claims(on_claim)
+            except Exception:
+                log.exception("claim watch unavailable; polling only")
         self._thread = threading.Thread(
             target=self.run, name="controller-reconcile", daemon=True
         )
@@ -176,5 +194,8 @@ class ControllerManager:
 
     def stop(self) -> None:
         self._stop.set()
+        self._kick.set()
+        if self._watch is not None:
+            self._watch.stop()
         if self._thread:
             self._thread.join(timeout=5)

@@ -45,6 +45,14 @@ class KubeClient(abc.ABC):
     def create_event(self, namespace: str, event: dict) -> None:
         """Emit a Kubernetes Event (operator visibility); best-effort."""
 
+    def watch_resource_claims(self, handler: Callable[[str, dict], None]):
+        """Subscribe to ResourceClaim changes: ``handler(event_type, obj)``
+        with event_type ADDED|MODIFIED|DELETED. Returns an object with
+        ``stop()``, or None if watching is unsupported (callers fall back
+        to polling) — the client-go informer analog (reference
+        imex.go:222-295)."""
+        return None
+
     @abc.abstractmethod
     def create_resource_slice(self, obj: dict) -> dict: ...
 
@@ -92,9 +100,11 @@ class InMemoryKube(KubeClient):
     def put_resource_claim(self, obj: dict) -> dict:
         with self._lock:
             key = f"{obj['metadata'].get('namespace','default')}/{obj['metadata']['name']}"
+            verb = "MODIFIED" if key in self.resource_claims else "ADDED"
             obj["metadata"]["resourceVersion"] = self._next_rv()
             self.resource_claims[key] = copy.deepcopy(obj)
-            return obj
+        self._emit("ResourceClaim", verb, obj)
+        return obj
 
     def put_node(self, obj: dict) -> dict:
         with self._lock:
@@ -138,6 +148,21 @@ class InMemoryKube(KubeClient):
     def create_event(self, namespace: str, event: dict) -> None:
         with self._lock:
             self.events.append({"namespace": namespace, **copy.deepcopy(event)})
+
+    def watch_resource_claims(self, handler: Callable[[str, dict], None]):
+        def hook(kind: str, verb: str, obj: dict) -> None:
+            if kind == "ResourceClaim" and verb in ("ADDED", "MODIFIED", "DELETED"):
+                handler(verb, copy.deepcopy(obj))
+
+        self.watchers.append(hook)
+        kube = self
+
+        class _Watch:
+            def stop(self_inner) -> None:
+                if hook in kube.watchers:
+                    kube.watchers.remove(hook)
+
+        return _Watch()
 
     def create_resource_slice(self, obj: dict) -> dict:
         with self._lock:

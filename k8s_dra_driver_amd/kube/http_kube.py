@@ -162,6 +162,46 @@ class HttpKube(KubeClient):
         out = self._req("GET", f"{RESOURCE_V1BETA1}/deviceclasses")
         return out.get("items", [])
 
+    def watch_resource_claims(self, handler):
+        """Streaming watch (?watch=true, JSON-line events) with reconnect —
+        the informer analog. stop() is lazy: the reader exits at the next
+        event or when the connection drops."""
+        stop_event = threading.Event()
+        client = self._client
+
+        def run():
+            while not stop_event.is_set():
+                try:
+                    with client.stream(
+                        "GET",
+                        f"{RESOURCE_V1BETA1}/resourceclaims?watch=true",
+                        timeout=httpx.Timeout(5.0, read=None),
+                    ) as r:
+                        for line in r.iter_lines():
+                            if stop_event.is_set():
+                                return
+                            line = line.strip()
+                            if not line:
+                                continue
+                            try:
+                                ev = json.loads(line)
+                            except json.JSONDecodeError:
+                                continue
+                            handler(ev.get("type", ""), ev.get("object") or {})
+                except Exception:
+                    if stop_event.is_set():
+                        return
+                    time.sleep(1.0)  # transient-error retry (imex.go parity)
+
+        thread = threading.Thread(target=run, name="claims-watch", daemon=True)
+        thread.start()
+
+        class _Watch:
+            def stop(self_inner) -> None:
+                stop_event.set()
+
+        return _Watch()
+
     def create_event(self, namespace: str, event: dict) -> None:
         try:
             self._req("POST", f"{CORE_V1}/namespaces/{namespace}/events", event)

@@ -50,6 +50,8 @@ class MiniApiServer:
 
             def do_GET(self):
                 path = self.path.split("?")[0]
+                if path == _CLAIMS_ALL and "watch=true" in self.path:
+                    return self._watch_claims()
                 m = _CLAIM_RE.match(path)
                 if m:
                     try:
@@ -78,6 +80,40 @@ class MiniApiServer:
                     except NotFound:
                         return self._json(404, {})
                 self._json(404, {"path": path})
+
+            def _watch_claims(self):
+                """Streamed watch: JSON-line events until client disconnect
+                (read-until-close framing)."""
+                import queue
+
+                q: "queue.Queue" = queue.Queue()
+
+                def hook(kind, verb, obj):
+                    if kind == "ResourceClaim" and verb in (
+                        "ADDED",
+                        "MODIFIED",
+                        "DELETED",
+                    ):
+                        q.put({"type": verb, "object": obj})
+
+                outer.store.watchers.append(hook)
+                try:
+                    self.send_response(200)
+                    self.send_header("Content-Type", "application/json")
+                    self.send_header("Connection", "close")
+                    self.end_headers()
+                    while True:
+                        try:
+                            ev = q.get(timeout=0.5)
+                        except queue.Empty:
+                            continue
+                        self.wfile.write((json.dumps(ev) + "\n").encode())
+                        self.wfile.flush()
+                except (BrokenPipeError, ConnectionResetError):
+                    pass
+                finally:
+                    if hook in outer.store.watchers:
+                        outer.store.watchers.remove(hook)
 
             def do_POST(self):
                 path = self.path.split("?")[0]

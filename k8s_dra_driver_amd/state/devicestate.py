@@ -171,7 +171,7 @@ class DeviceState:
                     missing,
                 )
             for dev in pc.devices:
-                if dev.parent_gpu_index >= 0:
+                if dev.parent_gpu_index >= 0 and not dev.admin:
                     self._gpu_holders.setdefault(dev.parent_gpu_index, set()).add(uid)
             # Re-register shared sessions so CU-range bookkeeping survives
             # a plugin restart (stale-state gap the reference leaves open,
@@ -222,10 +222,14 @@ class DeviceState:
 
             results, configs = self._parse_allocation(claim)
             prepared = self._prepare_devices(info, results, configs)
-            self.checkpoints.write(prepared)
-            fut = getattr(prepared, "_cdi_write", None)
-            if fut is not None:
-                fut.result()  # join the overlapped claim-spec write
+            try:
+                self.checkpoints.write(prepared)
+                fut = getattr(prepared, "_cdi_write", None)
+                if fut is not None:
+                    fut.result()  # join the overlapped claim-spec write
+            except BaseException:
+                self._rollback(prepared, info.uid)
+                raise
             with self._registry_lock:
                 for dev in prepared.devices:
                     # adminAccess (monitoring) claims neither block
@@ -317,8 +321,21 @@ class DeviceState:
             self.refresh_allocatable()
             self.write_base_cdi_spec()
 
+        try:
+            return self._prepare_after_partition(info, results, configs, prepared)
+        except BaseException:
+            # a failure past a successful mode switch must not leak it
+            self._rollback(prepared, info.uid)
+            raise
+
+    def _prepare_after_partition(
+        self,
+        info: _ClaimInfo,
+        results: List[dict],
+        configs: List[OpaqueConfig],
+        prepared: PreparedClaim,
+    ) -> PreparedClaim:
         # --- group results by sharing config -------------------------------
-        claim_devices: List[AllocatableDevice] = []
         per_result_dev: List[tuple] = []
         for r in results:
             name = r.get("device", "")
@@ -334,10 +351,8 @@ class DeviceState:
                     )
                 for p in parts:
                     per_result_dev.append((r, p))
-                    claim_devices.append(p)
             else:
                 per_result_dev.append((r, dev))
-                claim_devices.append(dev)
 
         # --- sharing --------------------------------------------------------
         shared_edits: Optional[ContainerEdits] = None
@@ -391,8 +406,7 @@ class DeviceState:
                     ts_devices, settings
                 )
             except Exception as e:
-                if prepared.shared_session_id and self.shared_manager:
-                    self.shared_manager.stop_session(prepared.shared_session_id)
+                # _rollback (invoked by the caller) undoes the session
                 raise PrepareError(f"time-slicing failed: {e}") from e
             if not prepared.sharing_strategy:
                 prepared.sharing_strategy = TIME_SLICING
@@ -433,6 +447,41 @@ class DeviceState:
                 )
             )
         return prepared
+
+    def _rollback(self, prepared: PreparedClaim, claim_uid: str) -> None:
+        """Best-effort undo of a partially-prepared claim: stop the shared
+        session, restore time-slice defaults, revert any mode switch, and
+        drop on-disk artifacts — so a failed Prepare leaves the node as it
+        found it (kubelet will retry from scratch)."""
+        try:
+            if prepared.shared_session_id and self.shared_manager:
+                self.shared_manager.stop_session(prepared.shared_session_id)
+            if prepared.timeslice_gpus:
+                self.ts_manager.restore_default(prepared.timeslice_gpus)
+            reverted = False
+            for gpu_index_s, modes in prepared.repartitioned.items():
+                try:
+                    if self.partition_manager.ensure_mode(
+                        int(gpu_index_s),
+                        modes[0],
+                        modes[1],
+                        requesting_claim=claim_uid,
+                        allow_dynamic=True,
+                    ):
+                        reverted = True
+                except RepartitionRefused as e:
+                    log.warning(
+                        "rollback: leaving gpu-%s partitioned (%s)",
+                        gpu_index_s,
+                        e,
+                    )
+            if reverted:
+                self.refresh_allocatable()
+                self.write_base_cdi_spec()
+            self.cdi.delete_claim_spec(claim_uid)
+            self.checkpoints.delete(claim_uid)
+        except Exception:
+            log.exception("rollback of claim %s incomplete", claim_uid)
 
     def _ensure_claim_spec(self, pc: PreparedClaim) -> None:
         """Regenerate the claim CDI spec if a crash between the overlapped

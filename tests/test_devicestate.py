@@ -306,3 +306,26 @@ class TestRecovery:
             t.join()
         assert not errs
         assert len(state.checkpoints.list_all()) == 8
+
+
+class TestOtherPartitionModes:
+    def test_dpx_and_qpx_prepare(self, tmp_path):
+        """DPX (2x4 XCD) and QPX (4x2 XCD) carve + 1:1 partition claims."""
+        state, lib = make_state(tmp_path)
+        state.partition_manager.ensure_mode(0, "DPX", "NPS1", allow_dynamic=True)
+        state.partition_manager.ensure_mode(1, "QPX", "NPS4", allow_dynamic=True)
+        state.refresh_allocatable()
+        names = {d.canonical_name for d in state.allocatable_devices()}
+        assert {"gpu-0-dpx-0", "gpu-0-dpx-1"} <= names
+        assert {"gpu-1-qpx-0", "gpu-1-qpx-3"} <= names
+        devs = state.prepare(make_claim("uid-dpx", ["gpu-0-dpx-1"]))
+        assert devs[0]["device_name"] == "gpu-0-dpx-1"
+        devs = state.prepare(make_claim("uid-qpx", ["gpu-1-qpx-2"]))
+        assert devs[0]["device_name"] == "gpu-1-qpx-2"
+        # capacities reflect the mode
+        qpx = next(
+            d for d in state.allocatable_devices()
+            if d.canonical_name == "gpu-1-qpx-2"
+        ).to_device()
+        assert qpx["basic"]["capacity"]["gpu.amd.com/computeUnits"]["value"] == "64"
+        assert qpx["basic"]["capacity"]["gpu.amd.com/memory"]["value"] == "72Gi"

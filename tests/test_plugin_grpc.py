@@ -249,3 +249,29 @@ def test_batch_prepare_is_concurrent(stack):
         c.namespace, c.name, c.uid = "default", f"claim-uid-b{i}", f"uid-b{i}"
     resp = prepare(req, timeout=20)
     assert all(resp.claims[f"uid-b{i}"].error == "" for i in range(8))
+
+
+def test_malformed_request_bytes_do_not_crash_server(stack):
+    """Garbage on the wire must yield a gRPC error, not kill the plugin."""
+    lib, kube, driver, server, channel = stack
+    m = V1BETA1
+    raw = channel.unary_unary(
+        f"/{m.service_name}/NodePrepareResources",
+        request_serializer=lambda b: b,  # send raw bytes
+        response_deserializer=lambda b: b,
+    )
+    with pytest.raises(grpc.RpcError):
+        raw(b"\xff\xfe\xfd this is not protobuf \x00\x01", timeout=5)
+    # server still alive and serving
+    put_claim(kube, "uid-after", ["gpu-4"])
+    prepare = _stub(
+        channel,
+        m,
+        "NodePrepareResources",
+        m.NodePrepareResourcesRequest,
+        m.NodePrepareResourcesResponse,
+    )
+    req = m.NodePrepareResourcesRequest()
+    c = req.claims.add()
+    c.namespace, c.name, c.uid = "default", "claim-uid-after", "uid-after"
+    assert prepare(req, timeout=10).claims["uid-after"].error == ""

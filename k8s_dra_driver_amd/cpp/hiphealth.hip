@@ -20,6 +20,8 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <algorithm>
+#include <cmath>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -253,6 +255,112 @@ double mfma_tflops(int device, int iters, int blocks) {
 }
 
 // ---------------------------------------------------------------------------
+// n-body (the reference's demo workload, gpu-test5.yaml:57 runs NVIDIA's
+// nbody sample with --benchmark; this is the gfx950-native equivalent):
+// all-pairs gravitation, bodies tiled through LDS so each position is
+// fetched from HBM once per tile instead of once per pair.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) nbody_step(
+    const float4* __restrict__ pos_in, float4* __restrict__ pos_out,
+    float4* __restrict__ vel, int n, float dt, float softening2) {
+  __shared__ float4 tile[256];
+  int gid = blockIdx.x * blockDim.x + threadIdx.x;
+  float4 my = pos_in[gid < n ? gid : 0];
+  float ax = 0.f, ay = 0.f, az = 0.f;
+  for (int base = 0; base < n; base += 256) {
+    int j = base + threadIdx.x;
+    tile[threadIdx.x] = pos_in[j < n ? j : 0];
+    __syncthreads();
+    int limit = min(256, n - base);
+#pragma unroll 8
+    for (int k = 0; k < limit; ++k) {
+      float4 other = tile[k];
+      float dx = other.x - my.x;
+      float dy = other.y - my.y;
+      float dz = other.z - my.z;
+      float r2 = dx * dx + dy * dy + dz * dz + softening2;
+      float inv_r = __frsqrt_rn(r2);
+      float f = other.w * inv_r * inv_r * inv_r;  // m / r^3
+      ax = fmaf(f, dx, ax);
+      ay = fmaf(f, dy, ay);
+      az = fmaf(f, dz, az);
+    }
+    __syncthreads();
+  }
+  if (gid < n) {
+    float4 v = vel[gid];
+    v.x = fmaf(ax, dt, v.x);
+    v.y = fmaf(ay, dt, v.y);
+    v.z = fmaf(az, dt, v.z);
+    vel[gid] = v;
+    my.x = fmaf(v.x, dt, my.x);
+    my.y = fmaf(v.y, dt, my.y);
+    my.z = fmaf(v.z, dt, my.z);
+    pos_out[gid] = my;
+  }
+}
+
+py::dict nbody_benchmark(int device, int num_bodies, int iters) {
+  HIP_CHECK(hipSetDevice(device));
+  int n = ((num_bodies + 255) / 256) * 256;
+  size_t bytes = (size_t)n * sizeof(float4);
+  float4 *pos_a = nullptr, *pos_b = nullptr, *vel = nullptr;
+  HIP_CHECK(hipMalloc(&pos_a, bytes));
+  HIP_CHECK(hipMalloc(&pos_b, bytes));
+  HIP_CHECK(hipMalloc(&vel, bytes));
+  // deterministic pseudo-random init on host
+  std::vector<float4> host(n);
+  unsigned s = 0x5a1ad;
+  for (int i = 0; i < n; ++i) {
+    auto rnd = [&s]() {
+      s = s * 1664525u + 1013904223u;
+      return (float)(s >> 8) / (float)(1u << 24) - 0.5f;
+    };
+    host[i] = make_float4(rnd(), rnd(), rnd(), 1.0f / n);
+  }
+  HIP_CHECK(hipMemcpy(pos_a, host.data(), bytes, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemset(vel, 0, bytes));
+  dim3 block(256), grid(n / 256);
+  const float dt = 1e-3f, soft2 = 1e-4f;
+  hipLaunchKernelGGL(nbody_step, grid, block, 0, 0, pos_a, pos_b, vel, n,
+                     dt, soft2);  // warmup
+  HIP_CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(nbody_step, grid, block, 0, 0, pos_a, pos_b, vel, n,
+                       dt, soft2);
+    std::swap(pos_a, pos_b);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  // sanity: positions remain finite
+  HIP_CHECK(hipMemcpy(host.data(), pos_a, bytes, hipMemcpyDeviceToHost));
+  bool finite = true;
+  for (int i = 0; i < n; i += 97)
+    if (!std::isfinite(host[i].x)) finite = false;
+  HIP_CHECK(hipFree(pos_a));
+  HIP_CHECK(hipFree(pos_b));
+  HIP_CHECK(hipFree(vel));
+  // ~20 flops per pair interaction (nbody convention), n^2 pairs per step
+  double gflops =
+      20.0 * (double)n * n * iters / ((double)ms * 1e-3) / 1e9;
+  py::dict out;
+  out["bodies"] = n;
+  out["iters"] = iters;
+  out["ms_total"] = ms;
+  out["gflops"] = gflops;
+  out["finite"] = finite;
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // burner for sharing demos
 // ---------------------------------------------------------------------------
 __global__ void valu_burn(float* sink, long long iters) {
@@ -320,4 +428,6 @@ PYBIND11_MODULE(_hiphealth, m) {
   m.def("mfma_tflops", &mfma_tflops, py::arg("device") = 0,
         py::arg("iters") = 8192, py::arg("blocks") = 2048);
   m.def("burn_ms", &burn_ms, py::arg("device") = 0, py::arg("millis") = 100);
+  m.def("nbody_benchmark", &nbody_benchmark, py::arg("device") = 0,
+        py::arg("num_bodies") = 65536, py::arg("iters") = 10);
 }

@@ -90,6 +90,13 @@ def main(argv=None) -> int:
     ap.add_argument("--benchmark", action="store_true", help="run bandwidth + MFMA probes")
     ap.add_argument("--burn-ms", type=int, default=0, help="occupancy burn duration")
     ap.add_argument(
+        "--nbody",
+        type=int,
+        default=0,
+        help="run the all-pairs n-body benchmark with this many bodies "
+        "(the reference demo's nbody --benchmark analog)",
+    )
+    ap.add_argument(
         "--allreduce-mb",
         type=int,
         default=0,
@@ -102,7 +109,7 @@ def main(argv=None) -> int:
     if not info["kfd"] or not info["render_nodes"]:
         print("ERROR: no GPU devices injected", file=sys.stderr)
         return 1
-    if args.benchmark or args.burn_ms:
+    if args.benchmark or args.burn_ms or args.nbody:
         from k8s_dra_driver_amd import _hiphealth
 
         n = _hiphealth.device_count()
@@ -115,6 +122,12 @@ def main(argv=None) -> int:
             if args.burn_ms:
                 ms = _hiphealth.burn_ms(d, args.burn_ms)
                 print(f"device {d}: burned {ms:.0f} ms")
+            if args.nbody:
+                r = _hiphealth.nbody_benchmark(d, args.nbody, 10)
+                print(
+                    f"device {d}: nbody {r['bodies']} bodies x {r['iters']} "
+                    f"iters = {r['gflops']:.0f} GFLOP/s finite={r['finite']}"
+                )
     if args.allreduce_mb:
         return run_allreduce(args.allreduce_mb)
     return 0

@@ -207,3 +207,15 @@ class TestTorchInterop:
         (x @ x).sum().backward()
         torch.cuda.synchronize()
         assert x.grad is not None
+
+
+class TestNbodyDemo:
+    def test_nbody_benchmark(self):
+        """The reference demo's nbody --benchmark analog on gfx950."""
+        from k8s_dra_driver_amd import _hiphealth
+
+        r = _hiphealth.nbody_benchmark(0, 65536, 10)
+        assert r["finite"], r
+        # 256 CUs of fp32 VALU: an LDS-tiled all-pairs kernel should clear
+        # 10 TFLOP/s comfortably (peak vector fp32 is ~157 TF)
+        assert r["gflops"] > 10_000, r

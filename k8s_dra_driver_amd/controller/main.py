@@ -27,6 +27,12 @@ def build_parser() -> argparse.ArgumentParser:
         "--metrics-port", type=int, default=int(_env("METRICS_PORT", "8085"))
     )
     p.add_argument(
+        "--diag-port",
+        type=int,
+        default=int(_env("DIAG_PORT", "0")),
+        help="serve /healthz + /debug diagnostics on this port (0 = off)",
+    )
+    p.add_argument(
         "--allocate-claims",
         default=_env("ALLOCATE_CLAIMS", "false").lower() == "true",
         action="store_true",
@@ -67,6 +73,12 @@ def main(argv=None) -> int:
     metrics = PluginMetrics()
     if args.metrics_port:
         metrics.serve(args.metrics_port)
+    diag = None
+    if args.diag_port:
+        from ..utils.diag import DiagServer
+
+        diag = DiagServer(args.diag_port)
+        diag.start()
 
     mgr = ControllerManager(
         kube,
@@ -80,6 +92,8 @@ def main(argv=None) -> int:
         signal.signal(sig, lambda *_: stop.set())
     log.info("amd-dra-controller ready")
     stop.wait()
+    if diag is not None:
+        diag.stop()
     mgr.stop()
     return 0
 

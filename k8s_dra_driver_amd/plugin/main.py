@@ -75,6 +75,12 @@ def build_parser() -> argparse.ArgumentParser:
         "--kube-api-burst", type=int, default=int(_env("KUBE_API_BURST", "100"))
     )
     p.add_argument(
+        "--diag-port",
+        type=int,
+        default=int(_env("DIAG_PORT", "0")),
+        help="serve /healthz + /debug diagnostics on this port (0 = off)",
+    )
+    p.add_argument(
         "--metrics-port",
         type=int,
         default=int(_env("METRICS_PORT", "0")),
@@ -113,6 +119,12 @@ def main(argv=None) -> int:
     metrics = PluginMetrics()
     if args.metrics_port:
         metrics.serve(args.metrics_port)
+    diag = None
+    if args.diag_port:
+        from ..utils.diag import DiagServer
+
+        diag = DiagServer(args.diag_port)
+        diag.start()
     driver = Driver(
         lib,
         kube,
@@ -148,6 +160,8 @@ def main(argv=None) -> int:
         "amd-dra-kubeletplugin ready: node=%s hal=%s", args.node_name, args.hal
     )
     stop.wait()
+    if diag is not None:
+        diag.stop()
     driver.health.stop()
     server.stop()
     driver.shutdown()

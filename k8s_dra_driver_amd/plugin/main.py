@@ -142,6 +142,7 @@ def main(argv=None) -> int:
         registry_dir=args.plugin_registration_path,
     )
     server.start()
+    server.start_socket_watchdog()
 
     # periodic orphan cleanup (reference TODO parity, driver.go:156-168)
     stop = threading.Event()

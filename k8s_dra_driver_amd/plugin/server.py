@@ -13,6 +13,7 @@ from __future__ import annotations
 
 import logging
 import os
+import threading
 from concurrent.futures import ThreadPoolExecutor
 from typing import List, Optional
 
@@ -138,6 +139,8 @@ class PluginServer:
         )
         self._server: Optional[grpc.Server] = None
         self._grpc_workers = grpc_workers
+        self._watchdog_stop = threading.Event()
+        self._watchdog: Optional[threading.Thread] = None
 
     def start(self) -> None:
         os.makedirs(self.plugin_dir, exist_ok=True)
@@ -169,7 +172,44 @@ class PluginServer:
         self._server = server
         log.info("DRA plugin serving on %s", self.plugin_sock)
 
+    def start_socket_watchdog(self, interval_s: float = 10.0) -> None:
+        """Re-bind when kubelet wipes the plugin/registry sockets (e.g. a
+        kubelet restart recreating plugins_registry). The vendored
+        kubeletplugin gets this from the registration protocol retry; here
+        a stat loop restarts the gRPC server on socket loss — prepared
+        claims are unaffected (state is on disk + in DeviceState)."""
+
+        def run():
+            while not self._watchdog_stop.wait(interval_s):
+                missing = [
+                    s
+                    for s in filter(
+                        None, [self.plugin_sock, self.registry_sock]
+                    )
+                    if not os.path.exists(s)
+                ]
+                if missing and self._server is not None:
+                    log.warning(
+                        "socket(s) %s vanished (kubelet restart?); "
+                        "re-binding",
+                        missing,
+                    )
+                    try:
+                        self._server.stop(grace=1.0).wait()
+                        self._server = None
+                        self.start()
+                    except Exception:
+                        log.exception("socket re-bind failed; will retry")
+
+        self._watchdog = threading.Thread(
+            target=run, name="socket-watchdog", daemon=True
+        )
+        self._watchdog.start()
+
     def stop(self, grace: float = 2.0) -> None:
+        self._watchdog_stop.set()
+        if self._watchdog is not None:
+            self._watchdog.join(timeout=5)
         if self._server is not None:
             self._server.stop(grace).wait()
             self._server = None

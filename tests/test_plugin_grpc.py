@@ -275,3 +275,26 @@ def test_malformed_request_bytes_do_not_crash_server(stack):
     c = req.claims.add()
     c.namespace, c.name, c.uid = "default", "claim-uid-after", "uid-after"
     assert prepare(req, timeout=10).claims["uid-after"].error == ""
+
+
+def test_socket_watchdog_rebinds(stack):
+    """kubelet-restart resilience: a wiped socket is re-bound."""
+    import os
+    import time
+
+    lib, kube, driver, server, channel = stack
+    server.start_socket_watchdog(interval_s=0.2)
+    os.unlink(server.registry_sock)
+    t0 = time.time()
+    while not os.path.exists(server.registry_sock) and time.time() - t0 < 10:
+        time.sleep(0.1)
+    assert os.path.exists(server.registry_sock), "socket not re-bound"
+    # and it serves again
+    reg_channel = grpc.insecure_channel(f"unix://{server.registry_sock}")
+    info = reg_channel.unary_unary(
+        f"/{REGISTRATION.service_name}/GetInfo",
+        request_serializer=lambda m: m.SerializeToString(),
+        response_deserializer=REGISTRATION.PluginInfo.FromString,
+    )(REGISTRATION.InfoRequest(), timeout=5)
+    assert info.name == DRIVER_NAME
+    reg_channel.close()

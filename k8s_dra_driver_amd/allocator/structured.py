@@ -321,6 +321,18 @@ class Allocator:
         results = self.allocate(
             claim.get("spec", {}), devices, pool=pool, in_use=in_use
         )
+        return self.attach_allocation(claim, results, node_name=node_name)
+
+    def attach_allocation(
+        self,
+        claim: dict,
+        results: List[AllocationResult],
+        *,
+        node_name: Optional[str] = None,
+    ) -> dict:
+        """Write status.allocation from precomputed results (lets callers
+        that already searched — e.g. the controller scheduler scoring
+        across nodes — avoid a second search)."""
         # DeviceClass-attached config merges in first with source=FromClass
         # (lower precedence than claim config — the kube-scheduler behavior
         # the reference consumes, device_state.go:462-477).

@@ -154,13 +154,5 @@ class ClaimScheduler:
         if best is None:
             return None
         _, node, results = best
-        pool = nodes[node][0]
-        # write the allocation into the claim object (status update follows)
-        alloc.allocate_into_claim(
-            claim,
-            nodes[node][1],
-            pool=pool,
-            in_use=in_use.get(pool, set()),
-            node_name=node,
-        )
+        alloc.attach_allocation(claim, results, node_name=node)
         return node, results

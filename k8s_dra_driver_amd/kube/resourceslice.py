@@ -39,11 +39,16 @@ class ResourceSlicePublisher:
         driver_name: str,
         node_name: str,
         pool_name: Optional[str] = None,
+        node_uid: str = "",
     ):
         self.client = client
         self.driver_name = driver_name
         self.node_name = node_name
         self.pool_name = pool_name or node_name
+        # With a node UID, slices carry an ownerReference so apiserver GC
+        # deletes them when the Node object goes away (the k8s-native
+        # complement to the controller's orphan sweep).
+        self.node_uid = node_uid
         self._lock = threading.Lock()
         self._generation = 0
         self._last_fingerprint: Optional[str] = None
@@ -77,10 +82,20 @@ class ResourceSlicePublisher:
             out = []
             for i, chunk in enumerate(chunks):
                 name = self._slice_name(i)
+                meta: dict = {"name": name}
+                if self.node_uid:
+                    meta["ownerReferences"] = [
+                        {
+                            "apiVersion": "v1",
+                            "kind": "Node",
+                            "name": self.node_name,
+                            "uid": self.node_uid,
+                        }
+                    ]
                 obj = {
                     "apiVersion": API_VERSION,
                     "kind": "ResourceSlice",
-                    "metadata": {"name": name},
+                    "metadata": meta,
                     "spec": {
                         "driver": self.driver_name,
                         "nodeName": self.node_name,

@@ -84,8 +84,18 @@ class Driver:
             ts_manager=TimeSlicingManager(lib),
             shared_manager=shared,
         )
+        node_uid = ""
+        try:
+            node_uid = (self.kube.get_node(node_name).get("metadata") or {}).get(
+                "uid", ""
+            )
+        except Exception:
+            pass  # no Node object (tests/bench) -> no owner reference
         self.publisher = ResourceSlicePublisher(
-            kube, driver_name=DRIVER_NAME, node_name=node_name
+            kube,
+            driver_name=DRIVER_NAME,
+            node_name=node_name,
+            node_uid=node_uid,
         )
         self._pool = ThreadPoolExecutor(
             max_workers=max_concurrent_claims, thread_name_prefix="claim"

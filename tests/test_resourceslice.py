@@ -71,3 +71,42 @@ def test_unpublish_all():
     pub.publish([_dev("gpu-0")])
     pub.unpublish_all()
     assert kube.list_resource_slices("gpu.amd.com") == []
+
+
+def test_owner_reference_when_node_uid_known():
+    kube = InMemoryKube()
+    pub = ResourceSlicePublisher(
+        kube, driver_name="gpu.amd.com", node_name="node-a", node_uid="node-uid-1"
+    )
+    pub.publish([_dev("gpu-0")])
+    meta = kube.list_resource_slices("gpu.amd.com")[0]["metadata"]
+    assert meta["ownerReferences"][0] == {
+        "apiVersion": "v1",
+        "kind": "Node",
+        "name": "node-a",
+        "uid": "node-uid-1",
+    }
+
+
+def test_driver_shutdown_unpublishes(tmp_path):
+    from k8s_dra_driver_amd.hal import FakeDeviceLib
+    from k8s_dra_driver_amd.plugin.driver import Driver
+
+    lib = FakeDeviceLib()
+    lib.open()
+    kube = InMemoryKube()
+    kube.put_node({"metadata": {"name": "n", "uid": "nuid"}})
+    driver = Driver(
+        lib,
+        kube,
+        node_name="n",
+        cdi_root=str(tmp_path / "cdi"),
+        checkpoint_root=str(tmp_path / "state"),
+        use_tmpfs=False,
+    )
+    driver.startup()
+    slices = kube.list_resource_slices("gpu.amd.com")
+    assert slices
+    assert slices[0]["metadata"]["ownerReferences"][0]["uid"] == "nuid"
+    driver.shutdown()
+    assert kube.list_resource_slices("gpu.amd.com") == []

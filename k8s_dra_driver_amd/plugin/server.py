@@ -178,6 +178,7 @@ class PluginServer:
         kubeletplugin gets this from the registration protocol retry; here
         a stat loop restarts the gRPC server on socket loss — prepared
         claims are unaffected (state is on disk + in DeviceState)."""
+        self._watchdog_stop.clear()  # allow restart after a stop()
 
         def run():
             while not self._watchdog_stop.wait(interval_s):

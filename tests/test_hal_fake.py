@@ -147,3 +147,21 @@ def test_allocatable_device_union(fake_lib):
     assert part.canonical_name == "gpu-1-dpx-1"
     assert part.parent_gpu.uuid == gpus[1].uuid
     assert whole.to_device()["name"] == "gpu-0"
+
+
+def test_dpx_nps2_memory_domains(fake_lib):
+    """DPX+NPS2: 2 partitions over 2 domains, 1:1 binding."""
+    fake_lib.set_compute_partition(2, "DPX")
+    fake_lib.set_memory_partition(2, "NPS2")
+    g = fake_lib.enumerate()[2]
+    assert (g.compute_partition, g.memory_partition) == ("DPX", "NPS2")
+    assert len(g.partitions) == 2
+    prof = g.partitions[0].profile
+    assert prof.num_memory_domains == 2
+    assert [prof.memory_domain_of(i) for i in range(2)] == [0, 1]
+    assert prof.memory_mib_per_partition == 288 * 1024 // 2
+    from k8s_dra_driver_amd.hal.model import partition_to_device
+
+    dev = partition_to_device(g, g.partitions[1])
+    assert dev["basic"]["attributes"]["gpu.amd.com/memoryDomain"] == {"int": 1}
+    assert dev["basic"]["capacity"]["gpu.amd.com/memory"] == {"value": "144Gi"}

@@ -115,6 +115,11 @@ class AmdSmiDeviceLib(DeviceLib):
             memory_mode = head.get("memory_partition", "NPS1") or "NPS1"
             kfd_node = kfd_by_node.get(head.get("kfd_node_id", -1))
 
+            # Assumption (pinned in tests/test_amdsmi_hal.py): in non-SPX
+            # modes each partition processor reports its own VRAM share,
+            # so the physical total is the sum over members. Unverifiable
+            # on the round-1 pool (partition switching blocked); revisit
+            # against bare metal.
             vram_mb = int(head.get("vram_size_mb", 0))
             cu_total = sum(
                 int(m.get("num_compute_units", 0) or 0) for m in members

@@ -51,7 +51,16 @@ def _allreduce_worker(rank, world, nelem, results):
     dt = time.perf_counter() - t0
     # ring all-reduce moves 2*(n-1)/n of the buffer per GPU per iteration
     bytes_moved = 2 * (world - 1) / world * nelem * 2 * iters
-    ok = bool(torch.all(x == float(world) ** (iters + 3)).item()) if world > 1 else True
+    # each all-reduce multiplies by world; bf16 rounds non-power-of-two
+    # worlds, so compare with tolerance
+    expected = float(world) ** (iters + 3)
+    ok = bool(
+        torch.allclose(
+            x.float(),
+            torch.full_like(x, expected, dtype=torch.float32),
+            rtol=0.05,
+        )
+    )
     results[rank] = (bytes_moved / dt / 1e9, ok)
     dist.destroy_process_group()
 

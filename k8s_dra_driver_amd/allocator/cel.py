@@ -38,6 +38,25 @@ class CelError(Exception):
     pass
 
 
+class _CelErr:
+    """Evaluation-error VALUE (not an exception): CEL's ||/&& absorb
+    operand errors (true || error -> true), so runtime errors must flow
+    as values through the evaluator and only surface if they reach the
+    top. Parse errors still raise CelError immediately."""
+
+    __slots__ = ("msg",)
+
+    def __init__(self, msg: str):
+        self.msg = msg
+
+    def __repr__(self):
+        return f"celerror({self.msg!r})"
+
+
+def _is_err(v) -> bool:
+    return isinstance(v, _CelErr)
+
+
 _TOKEN_RE = re.compile(
     r"""
     (?P<ws>\s+)
@@ -152,23 +171,48 @@ class _Parser:
         v = self.or_()
         if self.peek()[0] != "eof":
             raise CelError(f"trailing tokens at {self.peek()[1]!r}")
+        if _is_err(v):
+            raise CelError(v.msg)
         return v
+
+    @staticmethod
+    def _as_bool(v):
+        """CEL logical operand: bool or error value."""
+        if _is_err(v) or isinstance(v, bool):
+            return v
+        return _CelErr(f"logical operand is {type(v).__name__}, not bool")
 
     def or_(self) -> Any:
-        v = self.and_()
+        first = self.and_()
+        if self.peek()[1] != "||":
+            return first  # not a logical expression: pass through untyped
+        operands = [self._as_bool(first)]
         while self.peek()[1] == "||":
             self.next()
-            rhs = self.and_()  # note: not short-circuit; selectors are pure
-            v = bool(v) or bool(rhs)
-        return v
+            operands.append(self._as_bool(self.and_()))
+        # CEL commutative-or: any true -> true; else any error -> error
+        if any(v is True for v in operands):
+            return True
+        for v in operands:
+            if _is_err(v):
+                return v
+        return False
 
     def and_(self) -> Any:
-        v = self.rel()
+        first = self.rel()
+        if self.peek()[1] != "&&":
+            return first  # not a logical expression: pass through untyped
+        operands = [self._as_bool(first)]
         while self.peek()[1] == "&&":
             self.next()
-            rhs = self.rel()
-            v = bool(v) and bool(rhs)
-        return v
+            operands.append(self._as_bool(self.rel()))
+        # CEL commutative-and: any false -> false; else any error -> error
+        if any(v is False for v in operands):
+            return False
+        for v in operands:
+            if _is_err(v):
+                return v
+        return True
 
     def rel(self) -> Any:
         v = self.add()
@@ -176,13 +220,17 @@ class _Parser:
         if op in ("==", "!=", "<", "<=", ">", ">=", "in"):
             self.next()
             rhs = self.add()
+            if _is_err(v):
+                return v
+            if _is_err(rhs):
+                return rhs
             if op == "==":
                 return v == rhs
             if op == "!=":
                 return v != rhs
             if op == "in":
                 if not isinstance(rhs, list):
-                    raise CelError("'in' requires a list on the right")
+                    return _CelErr("'in' requires a list on the right")
                 return v in rhs
             try:
                 if op == "<":
@@ -193,7 +241,7 @@ class _Parser:
                     return v > rhs
                 return v >= rhs
             except TypeError as e:
-                raise CelError(str(e)) from e
+                return _CelErr(str(e))
         return v
 
     def add(self) -> Any:
@@ -201,22 +249,34 @@ class _Parser:
         while self.peek()[1] in ("+", "-"):
             op = self.next()[1]
             rhs = self.unary()
+            if _is_err(v):
+                continue
+            if _is_err(rhs):
+                v = rhs
+                continue
             try:
                 v = v + rhs if op == "+" else v - rhs
             except TypeError as e:
-                raise CelError(str(e)) from e
+                v = _CelErr(str(e))
         return v
 
     def unary(self) -> Any:
         t = self.peek()[1]
         if t == "!":
             self.next()
-            return not bool(self.unary())
+            v = self.unary()
+            if _is_err(v):
+                return v
+            if not isinstance(v, bool):
+                return _CelErr(f"! on {type(v).__name__}")
+            return not v
         if t == "-":
             self.next()
             v = self.unary()
-            if not isinstance(v, (int, float)):
-                raise CelError("unary - on non-number")
+            if _is_err(v):
+                return v
+            if not isinstance(v, (int, float)) or isinstance(v, bool):
+                return _CelErr("unary - on non-number")
             return -v
         return self.postfix()
 
@@ -242,22 +302,34 @@ class _Parser:
                 return v
 
     def member(self, v: Any, name: str) -> Any:
+        if _is_err(v):
+            return v
         if isinstance(v, _AttrMap):
-            return v.get(name)
+            try:
+                return v.get(name)
+            except CelError as e:
+                return _CelErr(str(e))
         if isinstance(v, dict):
             if name not in v:
-                raise CelError(f"no field {name!r}")
+                return _CelErr(f"no field {name!r}")
             return v[name]
-        raise CelError(f"member access .{name} on {type(v).__name__}")
+        return _CelErr(f"member access .{name} on {type(v).__name__}")
 
     def index(self, v: Any, idx: Any) -> Any:
+        if _is_err(v):
+            return v
+        if _is_err(idx):
+            return idx
         if isinstance(v, dict):
             if idx not in v:
-                raise CelError(f"no key {idx!r}")
+                return _CelErr(f"no key {idx!r}")
             return v[idx]
         if isinstance(v, list):
-            return v[int(idx)]
-        raise CelError(f"indexing {type(v).__name__}")
+            try:
+                return v[int(idx)]
+            except (IndexError, ValueError, TypeError) as e:
+                return _CelErr(str(e))
+        return _CelErr(f"indexing {type(v).__name__}")
 
     def call_method(self, v: Any, name: str) -> Any:
         self.expect("(")
@@ -268,8 +340,13 @@ class _Parser:
                 self.next()
                 args.append(self.or_())
         self.expect(")")
+        if _is_err(v):
+            return v
+        for a in args:
+            if _is_err(a):
+                return a
         if not isinstance(v, str):
-            raise CelError(f"method .{name}() on {type(v).__name__}")
+            return _CelErr(f"method .{name}() on {type(v).__name__}")
         if name == "lowerAscii":
             return v.lower()
         if name == "upperAscii":
@@ -284,7 +361,7 @@ class _Parser:
             return v.startswith(str(args[0]))
         if name == "endsWith":
             return v.endswith(str(args[0]))
-        raise CelError(f"unknown method {name!r}")
+        return _CelErr(f"unknown method {name!r}")
 
     def primary(self) -> Any:
         kind, t = self.next()
@@ -316,10 +393,15 @@ class _Parser:
                 self.next()
                 arg = self.or_()
                 self.expect(")")
-                return Quantity(parse_quantity_bytes(str(arg)))
+                if _is_err(arg):
+                    return arg
+                try:
+                    return Quantity(parse_quantity_bytes(str(arg)))
+                except Exception as e:
+                    return _CelErr(f"quantity(): {e}")
             if t in self.env:
                 return self.env[t]
-            raise CelError(f"unknown identifier {t!r}")
+            return _CelErr(f"unknown identifier {t!r}")
         raise CelError(f"unexpected token {t!r}")
 
 

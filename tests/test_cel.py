@@ -130,3 +130,63 @@ class TestErrorSemantics:
     def test_non_bool_result_rejected(self):
         with pytest.raises(CelError):
             evaluate("device.attributes['gpu.amd.com'].index", GPU, "gpu.amd.com")
+
+
+class TestCelErrorAbsorption:
+    """Kubernetes CEL ||/&& are commutative error absorbers:
+    true || error -> true, false && error -> false. The evaluator must
+    match, or selectors that guard optional attributes reject devices the
+    real scheduler would accept."""
+
+    def test_true_or_error_is_true(self):
+        assert evaluate(
+            "device.attributes['gpu.amd.com'].type == 'gpu' || "
+            "device.attributes['gpu.amd.com'].missingAttr == 1",
+            GPU,
+            "gpu.amd.com",
+        )
+        # commutative: error first
+        assert evaluate(
+            "device.attributes['gpu.amd.com'].missingAttr == 1 || "
+            "device.attributes['gpu.amd.com'].type == 'gpu'",
+            GPU,
+            "gpu.amd.com",
+        )
+
+    def test_false_and_error_is_false(self):
+        assert not evaluate(
+            "device.attributes['gpu.amd.com'].type == 'cpu' && "
+            "device.attributes['gpu.amd.com'].missingAttr == 1",
+            GPU,
+            "gpu.amd.com",
+        )
+        assert not evaluate(
+            "device.attributes['gpu.amd.com'].missingAttr == 1 && "
+            "device.attributes['gpu.amd.com'].type == 'cpu'",
+            GPU,
+            "gpu.amd.com",
+        )
+
+    def test_unresolved_error_still_raises(self):
+        with pytest.raises(CelError):
+            evaluate(
+                "device.attributes['gpu.amd.com'].type == 'cpu' || "
+                "device.attributes['gpu.amd.com'].missingAttr == 1",
+                GPU,
+                "gpu.amd.com",
+            )
+        with pytest.raises(CelError):
+            evaluate(
+                "device.attributes['gpu.amd.com'].type == 'gpu' && "
+                "device.attributes['gpu.amd.com'].missingAttr == 1",
+                GPU,
+                "gpu.amd.com",
+            )
+
+    def test_non_bool_logical_operand_is_error(self):
+        with pytest.raises(CelError):
+            evaluate(
+                "device.attributes['gpu.amd.com'].productName && true",
+                GPU,
+                "gpu.amd.com",
+            )

@@ -102,6 +102,7 @@ class Driver:
         )
         # Republish whenever the allocatable set changes (repartition).
         self.state.on_allocatable_change = self.publish_resources
+        self.state.on_repartition = self.metrics.repartitions.inc
         # Failure detection: unhealthy GPUs are pulled from publication
         # (start()ed by main.py; tests drive check_once directly).
         from .health import HealthMonitor

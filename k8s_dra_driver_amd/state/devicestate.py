@@ -98,6 +98,8 @@ class DeviceState:
         #: optional callback fired after the allocatable set changes
         #: (repartition) — the Driver republishes ResourceSlices from it.
         self.on_allocatable_change = None
+        #: optional callback per performed mode switch (metrics)
+        self.on_repartition = None
 
         self._registry_lock = threading.Lock()
         self._claim_locks: Dict[str, threading.Lock] = {}
@@ -310,6 +312,8 @@ class DeviceState:
             except RepartitionRefused as e:
                 raise PrepareError(str(e)) from e
             if switched:
+                if self.on_repartition is not None:
+                    self.on_repartition()
                 prepared.repartitioned[str(gpu_index)] = [
                     cur.compute_partition,
                     cur.memory_partition,
@@ -561,6 +565,8 @@ class DeviceState:
                         allow_dynamic=True,
                     ):
                         restored = True
+                        if self.on_repartition is not None:
+                            self.on_repartition()
                 except RepartitionRefused as e:
                     log.warning(
                         "leaving gpu-%s partitioned (%s)", gpu_index_s, e

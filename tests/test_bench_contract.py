@@ -74,3 +74,64 @@ def test_metrics_histograms():
     assert "dra_prepare_seconds_bucket" in text
     assert "dra_prepared_claims_total 1.0" in text
     assert "dra_allocatable_devices 8.0" in text
+
+
+def test_repartition_metric_counts(tmp_path):
+    from k8s_dra_driver_amd.api.types import API_GROUP_VERSION
+    from k8s_dra_driver_amd.hal import FakeDeviceLib
+    from k8s_dra_driver_amd.kube.client import InMemoryKube
+    from k8s_dra_driver_amd.plugin.driver import ClaimRef, Driver
+    from prometheus_client import generate_latest
+
+    lib = FakeDeviceLib()
+    lib.open()
+    kube = InMemoryKube()
+    driver = Driver(
+        lib,
+        kube,
+        node_name="n",
+        cdi_root=str(tmp_path / "cdi"),
+        checkpoint_root=str(tmp_path / "s"),
+        use_tmpfs=False,
+    )
+    driver.startup()
+    kube.put_resource_claim(
+        {
+            "metadata": {"namespace": "d", "name": "c", "uid": "u"},
+            "status": {
+                "allocation": {
+                    "devices": {
+                        "results": [
+                            {
+                                "request": "g",
+                                "driver": "gpu.amd.com",
+                                "pool": "n",
+                                "device": "gpu-0",
+                            }
+                        ],
+                        "config": [
+                            {
+                                "source": "FromClaim",
+                                "requests": [],
+                                "opaque": {
+                                    "driver": "gpu.amd.com",
+                                    "parameters": {
+                                        "apiVersion": API_GROUP_VERSION,
+                                        "kind": "PartitionConfig",
+                                        "computePartition": "CPX",
+                                        "memoryPartition": "NPS1",
+                                        "allowDynamicRepartition": True,
+                                    },
+                                },
+                            }
+                        ],
+                    }
+                }
+            },
+        }
+    )
+    driver.node_prepare_resources([ClaimRef("d", "c", "u")])
+    driver.node_unprepare_resources([ClaimRef("d", "c", "u")])
+    text = generate_latest(driver.metrics.registry).decode()
+    # one switch at prepare + one restore at unprepare
+    assert "dra_repartitions_total 2.0" in text

@@ -124,7 +124,7 @@ def cmd_health(args) -> int:
 
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser("amd-dra-ctl")
-    ap.add_argument("--hal", default="amdsmi", choices=["amdsmi", "fake"])
+    ap.add_argument("--hal", default="amdsmi", choices=["amdsmi", "kfd", "fake"])
     sub = ap.add_subparsers(dest="cmd", required=True)
     sub.add_parser("list")
     sub.add_parser("topology")

@@ -17,6 +17,10 @@ def new_device_lib(backend: str = "auto", **kw):
     """
     if backend == "fake":
         return FakeDeviceLib(**kw)
+    if backend == "kfd":
+        from .kfd import KfdDeviceLib
+
+        return KfdDeviceLib(**kw)
     if backend in ("amdsmi", "auto"):
         from .amdsmi import AmdSmiDeviceLib
 

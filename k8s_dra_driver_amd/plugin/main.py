@@ -41,7 +41,7 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument(
         "--hal",
         default=_env("DRA_HAL", "amdsmi"),
-        choices=["amdsmi", "fake"],
+        choices=["amdsmi", "kfd", "fake"],
         help="hardware backend (fake = 8xMI355X model for dev clusters)",
     )
     p.add_argument(

@@ -82,3 +82,94 @@ def test_missing_tree_is_empty(tmp_path):
     topo = KfdTopology(str(tmp_path / "nope"))
     assert not topo.available()
     assert topo.nodes() == []
+
+
+class TestKfdOnlyBackend:
+    """Degraded-mode HAL over the fixture tree (no libamd_smi)."""
+
+    def _fixture_with_vram(self, root):
+        build_fixture(root)
+        nodes = os.path.join(root, "class", "kfd", "kfd", "topology", "nodes")
+        for i in (1, 2):
+            banks = os.path.join(nodes, str(i), "mem_banks", "0")
+            os.makedirs(banks)
+            with open(os.path.join(banks, "properties"), "w") as f:
+                f.write(f"heap_type 1\nsize_in_bytes {288 * 1024**3}\n")
+        return root
+
+    def test_enumerate(self, tmp_path):
+        from k8s_dra_driver_amd.hal.kfd import KfdDeviceLib
+
+        lib = KfdDeviceLib(str(self._fixture_with_vram(tmp_path)))
+        lib.open()
+        gpus = lib.enumerate()
+        assert len(gpus) == 2
+        g = gpus[0]
+        assert g.architecture == "gfx950"
+        assert g.vram_total_mib == 288 * 1024
+        assert g.cu_count == 256
+        assert g.render_minor == 128
+        assert g.uuid.startswith("kfd-")
+        assert gpus[0].xgmi_peer_oam_ids() == [1]
+
+    def test_partition_control_refused(self, tmp_path):
+        from k8s_dra_driver_amd.hal.base import HalNotSupported
+        from k8s_dra_driver_amd.hal.kfd import KfdDeviceLib
+
+        lib = KfdDeviceLib(str(self._fixture_with_vram(tmp_path)))
+        lib.open()
+        import pytest as _pytest
+
+        with _pytest.raises(HalNotSupported):
+            lib.set_compute_partition(0, "CPX")
+
+    def test_unavailable_without_kfd(self, tmp_path):
+        from k8s_dra_driver_amd.hal.base import HalUnavailable
+        from k8s_dra_driver_amd.hal.kfd import KfdDeviceLib
+
+        import pytest as _pytest
+
+        with _pytest.raises(HalUnavailable):
+            KfdDeviceLib(str(tmp_path / "empty")).open()
+
+    def test_full_driver_over_kfd_backend(self, tmp_path):
+        """The whole prepare path runs on the degraded backend."""
+        from k8s_dra_driver_amd.hal.kfd import KfdDeviceLib
+        from k8s_dra_driver_amd.kube.client import InMemoryKube
+        from k8s_dra_driver_amd.plugin.driver import ClaimRef, Driver
+
+        lib = KfdDeviceLib(str(self._fixture_with_vram(tmp_path / "sys")))
+        lib.open()
+        kube = InMemoryKube()
+        driver = Driver(
+            lib,
+            kube,
+            node_name="n",
+            cdi_root=str(tmp_path / "cdi"),
+            checkpoint_root=str(tmp_path / "state"),
+            use_tmpfs=False,
+        )
+        driver.startup()
+        devs = kube.list_resource_slices("gpu.amd.com")[0]["spec"]["devices"]
+        assert len(devs) == 2
+        kube.put_resource_claim(
+            {
+                "metadata": {"namespace": "d", "name": "c", "uid": "u"},
+                "status": {
+                    "allocation": {
+                        "devices": {
+                            "results": [
+                                {
+                                    "request": "g",
+                                    "driver": "gpu.amd.com",
+                                    "pool": "n",
+                                    "device": devs[0]["name"],
+                                }
+                            ]
+                        }
+                    }
+                },
+            }
+        )
+        res = driver.node_prepare_resources([ClaimRef("d", "c", "u")])["u"]
+        assert not res.error

@@ -219,3 +219,21 @@ class TestNbodyDemo:
         # 256 CUs of fp32 VALU: an LDS-tiled all-pairs kernel should clear
         # 10 TFLOP/s comfortably (peak vector fp32 is ~157 TF)
         assert r["gflops"] > 10_000, r
+
+
+class TestKfdBackendOnHardware:
+    def test_kfd_only_enumeration_matches_amdsmi(self, real_lib):
+        """Degraded backend agrees with amdsmi on the basics."""
+        from k8s_dra_driver_amd.hal.kfd import KfdDeviceLib
+
+        kfd = KfdDeviceLib()
+        kfd.open()
+        kfd_gpus = kfd.enumerate()
+        smi_gpus = real_lib.enumerate()
+        assert len(kfd_gpus) == len(smi_gpus)
+        k, s = kfd_gpus[0], smi_gpus[0]
+        assert k.architecture == s.architecture == "gfx950"
+        assert k.render_minor == s.render_minor
+        assert k.cu_count == s.cu_count == 256
+        # VRAM from mem_banks within 2% of amdsmi's number
+        assert abs(k.vram_total_mib - s.vram_total_mib) < s.vram_total_mib * 0.02

@@ -120,6 +120,7 @@ class Driver:
 
     def shutdown(self, unpublish: bool = True) -> None:
         self._pool.shutdown(wait=True)
+        self.state.close()
         if unpublish:
             try:
                 self.publisher.unpublish_all()

@@ -136,6 +136,10 @@ class DeviceState:
         if self.on_allocatable_change is not None:
             self.on_allocatable_change()
 
+    def close(self) -> None:
+        """Release the write pool (Driver.shutdown calls this)."""
+        self._write_pool.shutdown(wait=True)
+
     def allocatable_devices(self) -> List[AllocatableDevice]:
         with self._registry_lock:
             return list(self._allocatable.values())

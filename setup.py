@@ -36,6 +36,7 @@ setup(
             "amd-dra-controller=k8s_dra_driver_amd.controller.main:main",
             "amd-dra-ctl=k8s_dra_driver_amd.ctl:main",
             "amd-dra-workload=k8s_dra_driver_amd.workload:main",
+            "amd-dra-demo=k8s_dra_driver_amd.demo_runner:main",
         ]
     },
 )

@@ -226,6 +226,11 @@ class DemoHarness:
             uc.namespace, uc.name, uc.uid = "demo", name, uid
             unprepare(ureq, timeout=30)
         channel.close()
+        # release the claims so the next spec's allocation sees free
+        # devices (the kubelet+GC role at pod deletion)
+        with self.api.store._lock:
+            for name in claim_uid_by_key:
+                self.api.store.resource_claims.pop(f"demo/{name}", None)
         print(f"[demo] done rc={rc}")
         return rc
 

@@ -72,6 +72,14 @@ class ResourceSlicePublisher:
             }
             if fp == self._last_fingerprint and existing:
                 return list(existing.values())
+            if self._generation == 0 and existing:
+                # plugin restart: never regress the pool generation below
+                # what is already published (the scheduler treats higher
+                # generations as authoritative for multi-slice pools)
+                self._generation = max(
+                    (s["spec"].get("pool", {}).get("generation", 0))
+                    for s in existing.values()
+                )
             self._generation += 1
 
             chunks = [

@@ -110,3 +110,17 @@ def test_driver_shutdown_unpublishes(tmp_path):
     assert slices[0]["metadata"]["ownerReferences"][0]["uid"] == "nuid"
     driver.shutdown()
     assert kube.list_resource_slices("gpu.amd.com") == []
+
+
+def test_generation_survives_publisher_restart():
+    """A restarted plugin must not regress the pool generation."""
+    kube = InMemoryKube()
+    pub1 = _pub(kube)
+    pub1.publish([_dev("gpu-0")])
+    pub1.publish([_dev("gpu-0"), _dev("gpu-1")])  # generation 2
+    pub2 = _pub(kube)  # restart
+    pub2.publish([_dev("gpu-0")])  # changed set -> must bump past 2
+    gen = kube.list_resource_slices("gpu.amd.com")[0]["spec"]["pool"][
+        "generation"
+    ]
+    assert gen == 3

@@ -102,6 +102,66 @@ class DemoHarness:
         )
         print(f"[demo] system up: {n} device(s) published on demo-node")
 
+    def carve(self, gpu_name: str, compute: str, memory: str = "NPS1") -> None:
+        """Pre-carve a GPU via a held claim (enables the partition demos,
+        e.g. gpu-test4's four-partitions-one-die constraint)."""
+        uid = f"carve-{gpu_name}"
+        self.api.store.put_resource_claim(
+            {
+                "metadata": {"namespace": "demo", "name": uid, "uid": uid},
+                "status": {
+                    "allocation": {
+                        "devices": {
+                            "results": [
+                                {
+                                    "request": "gpu",
+                                    "driver": DRIVER_NAME,
+                                    "pool": "demo-node",
+                                    "device": gpu_name,
+                                }
+                            ],
+                            "config": [
+                                {
+                                    "source": "FromClaim",
+                                    "requests": [],
+                                    "opaque": {
+                                        "driver": DRIVER_NAME,
+                                        "parameters": {
+                                            "apiVersion": "resource.gpu.amd.com/v1alpha1",
+                                            "kind": "PartitionConfig",
+                                            "computePartition": compute,
+                                            "memoryPartition": memory,
+                                            "allowDynamicRepartition": True,
+                                        },
+                                    },
+                                }
+                            ],
+                        }
+                    }
+                },
+            }
+        )
+        channel, prepare, _ = self._grpc()
+        m = V1BETA1
+        req = m.NodePrepareResourcesRequest()
+        c = req.claims.add()
+        c.namespace, c.name, c.uid = "demo", uid, uid
+        resp = prepare(req, timeout=60)
+        channel.close()
+        if resp.claims[uid].error:
+            raise RuntimeError(f"carve failed: {resp.claims[uid].error}")
+        parts = [d.device_name for d in resp.claims[uid].devices]
+        print(f"[demo] carved {gpu_name} -> {len(parts)} partition(s) ({compute}/{memory})")
+        _wait(
+            lambda: any(
+                "-" in d["name"] and d["name"].startswith(gpu_name + "-")
+                for s in self.api.store.list_resource_slices(DRIVER_NAME)
+                for d in s["spec"]["devices"]
+            ),
+            30,
+            "partition republication",
+        )
+
     # -- kubelet role -------------------------------------------------------
     def _grpc(self):
         channel = grpc.insecure_channel(f"unix://{self.plugin_sock}")
@@ -250,12 +310,22 @@ def main(argv=None) -> int:
     ap = argparse.ArgumentParser("amd-dra-demo")
     ap.add_argument("specs", nargs="+", help="demo spec YAML file(s)")
     ap.add_argument("--hal", default="fake", choices=["fake", "amdsmi", "kfd"])
+    ap.add_argument(
+        "--carve",
+        action="append",
+        default=[],
+        metavar="GPU:MODE[:NPS]",
+        help="pre-carve a GPU before running specs, e.g. gpu-0:CPX:NPS4",
+    )
     ap.add_argument("-v", "--verbose", action="store_true")
     args = ap.parse_args(argv)
     harness = DemoHarness(hal=args.hal, verbose=args.verbose)
     rc = 0
     try:
         harness.start()
+        for spec in args.carve:
+            parts = spec.split(":")
+            harness.carve(parts[0], parts[1], parts[2] if len(parts) > 2 else "NPS1")
         for spec in args.specs:
             rc |= harness.run_spec(spec)
     finally:

@@ -14,11 +14,14 @@ SPECS = os.path.join(
 )
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(240)
 def test_demo_runner_end_to_end(capsys):
     rc = main(
         [
+            "--carve",
+            "gpu-7:CPX",
             os.path.join(SPECS, "gpu-test1.yaml"),
+            os.path.join(SPECS, "gpu-test4.yaml"),
             os.path.join(SPECS, "gpu-test5.yaml"),
             os.path.join(SPECS, "partition-carve.yaml"),
         ]
@@ -28,4 +31,11 @@ def test_demo_runner_end_to_end(capsys):
     assert "system up: 8 device(s)" in out
     assert "POD pod1: devices=['gpu-0']" in out
     assert "HSA_CU_MASK" in out  # shared-compute pod env
-    assert "gpu-2-cpx-7" in out or "cpx-7" in out  # dynamic carve
+    assert "cpx-" in out  # dynamic carve + pre-carved partitions
+    # gpu-test4: four partitions constrained to ONE parent die
+    import re
+
+    m = re.search(r"gpu-test4.*?devices=\[([^\]]+)\]", out, re.S)
+    assert m, out
+    parents = {d.strip(" '").rsplit("-", 2)[0] for d in m.group(1).split(",")}
+    assert len(parents) == 1

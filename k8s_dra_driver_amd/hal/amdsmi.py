@@ -108,7 +108,17 @@ class AmdSmiDeviceLib(DeviceLib):
             groups[key].append(p)
 
         out: List[GpuInfo] = []
-        for gpu_index, (key, members) in enumerate(sorted(groups.items(), key=lambda kv: str(kv[0]))):
+        # Order groups numerically first (int OAM ids), then lexically for
+        # string keys (UUID fallback), so gpu_index is stable across
+        # restarts on nodes with >=10 devices ('10' must not sort before
+        # '2' — indices are persisted in checkpoints as parent_gpu_index
+        # and drive holder/drain/partition bookkeeping).
+        for gpu_index, (key, members) in enumerate(
+            sorted(
+                groups.items(),
+                key=lambda kv: (isinstance(kv[0], str), kv[0]),
+            )
+        ):
             members.sort(key=lambda p: p.get("current_partition_id", 0))
             head = members[0]
             compute_mode = head.get("compute_partition", "SPX") or "SPX"

@@ -23,6 +23,7 @@ CDI claim spec -> checkpoint write.
 
 from __future__ import annotations
 
+import contextlib
 import logging
 import os
 import threading
@@ -56,6 +57,16 @@ DRIVER_NAME = "gpu.amd.com"
 
 class PrepareError(RuntimeError):
     pass
+
+
+class _RefLock:
+    """A lock plus a waiter/holder refcount (guarded by the registry lock)."""
+
+    __slots__ = ("lock", "refs")
+
+    def __init__(self) -> None:
+        self.lock = threading.Lock()
+        self.refs = 0
 
 
 @dataclass
@@ -102,7 +113,7 @@ class DeviceState:
         self.on_repartition = None
 
         self._registry_lock = threading.Lock()
-        self._claim_locks: Dict[str, threading.Lock] = {}
+        self._claim_locks: Dict[str, "_RefLock"] = {}
         #: overlaps the claim CDI-spec fsync with the checkpoint fsync
         self._write_pool = ThreadPoolExecutor(
             max_workers=4, thread_name_prefix="cdi-write"
@@ -187,12 +198,16 @@ class DeviceState:
 
                 spec_uids = set(self.cdi.list_claim_spec_uids())
                 if uid in spec_uids:
+                    # claim_env carries the checkpointed HSA_CU_MASK entries;
+                    # recover_session parses them back into CU-range
+                    # bookkeeping so a post-restart SharedCompute claim on the
+                    # same GPU cannot be handed an overlapping CU slice.
                     self.shared_manager.recover_session(
                         SharedSession(
                             session_id=pc.shared_session_id or uid[:36],
                             claim_uid=uid,
                             shm_dir="",
-                            env=[],
+                            env=list(pc.claim_env),
                             mounts=[],
                             gpu_indices=[
                                 d.parent_gpu_index for d in pc.devices
@@ -205,9 +220,26 @@ class DeviceState:
     # ------------------------------------------------------------------
     # locking
     # ------------------------------------------------------------------
-    def _claim_lock(self, uid: str) -> threading.Lock:
+    @contextlib.contextmanager
+    def _claim_lock(self, uid: str):
+        """Refcounted per-claim mutex.
+
+        Entries are GC'd only when no thread holds or waits on them, so a
+        kubelet retry racing an unprepare can never mint a second lock for
+        the same claim and run two operations concurrently (the
+        pop-while-waiters-blocked race of the old ``pop`` on unprepare).
+        """
         with self._registry_lock:
-            return self._claim_locks.setdefault(uid, threading.Lock())
+            rl = self._claim_locks.setdefault(uid, _RefLock())
+            rl.refs += 1
+        try:
+            with rl.lock:
+                yield
+        finally:
+            with self._registry_lock:
+                rl.refs -= 1
+                if rl.refs == 0 and self._claim_locks.get(uid) is rl:
+                    del self._claim_locks[uid]
 
     def claims_holding_gpu(self, gpu_index: int) -> List[str]:
         with self._registry_lock:
@@ -581,5 +613,3 @@ class DeviceState:
 
             self.cdi.delete_claim_spec(claim_uid)
             self.checkpoints.delete(claim_uid)
-            with self._registry_lock:
-                self._claim_locks.pop(claim_uid, None)

@@ -53,6 +53,12 @@ class KubeClient(abc.ABC):
         imex.go:222-295)."""
         return None
 
+    def watch_resource_slices(self, handler: Callable[[str, dict], None]):
+        """Subscribe to ResourceSlice changes (drift self-heal input,
+        reference resourceslicecontroller.go:407-431). Same contract as
+        :meth:`watch_resource_claims`; None if unsupported."""
+        return None
+
     @abc.abstractmethod
     def create_resource_slice(self, obj: dict) -> dict: ...
 
@@ -153,6 +159,23 @@ class InMemoryKube(KubeClient):
         def hook(kind: str, verb: str, obj: dict) -> None:
             if kind == "ResourceClaim" and verb in ("ADDED", "MODIFIED", "DELETED"):
                 handler(verb, copy.deepcopy(obj))
+
+        self.watchers.append(hook)
+        kube = self
+
+        class _Watch:
+            def stop(self_inner) -> None:
+                if hook in kube.watchers:
+                    kube.watchers.remove(hook)
+
+        return _Watch()
+
+    def watch_resource_slices(self, handler: Callable[[str, dict], None]):
+        verb_map = {"create": "ADDED", "update": "MODIFIED", "delete": "DELETED"}
+
+        def hook(kind: str, verb: str, obj: dict) -> None:
+            if kind == "ResourceSlice" and verb in verb_map:
+                handler(verb_map[verb], copy.deepcopy(obj))
 
         self.watchers.append(hook)
         kube = self

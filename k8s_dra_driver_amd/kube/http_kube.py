@@ -1,28 +1,36 @@
 """HTTP KubeClient: talks to a real apiserver over REST (httpx).
 
 The client-go analog of ``pkg/flags/kubeclient.go:30-107``: in-cluster
-service-account config or a kubeconfig file, with client-side QPS/burst
-rate limiting (reference flags kube-api-qps/kube-api-burst).
+service-account config or a kubeconfig file (full auth surface — inline
+cert data, client TLS, tokenFile, exec plugins — via :mod:`.auth`), with
+client-side QPS/burst rate limiting (reference flags kube-api-qps /
+kube-api-burst).
+
+Watches carry resourceVersion bookkeeping: each stream starts from a LIST
+(whose items are replayed to the handler as synthetic events, so consumers
+are level-triggered), then watches from that resourceVersion; reconnects
+resume from the last seen version, and a 410 Gone falls back to a fresh
+re-list — the client-go reflector contract
+(vendor ``resourceslicecontroller.go:407-431`` relies on it).
 """
 
 from __future__ import annotations
 
 import json
-import os
+import logging
 import threading
 import time
-from typing import Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
 import httpx
-import yaml
 
+from .auth import KubeConnection, load_in_cluster, load_kubeconfig
 from .client import Conflict, KubeClient, NotFound
+
+log = logging.getLogger(__name__)
 
 RESOURCE_V1BETA1 = "/apis/resource.k8s.io/v1beta1"
 CORE_V1 = "/api/v1"
-
-SA_TOKEN = "/var/run/secrets/kubernetes.io/serviceaccount/token"
-SA_CA = "/var/run/secrets/kubernetes.io/serviceaccount/ca.crt"
 
 
 class _RateLimiter:
@@ -55,62 +63,65 @@ class HttpKube(KubeClient):
         self,
         kubeconfig: Optional[str] = None,
         *,
+        context: Optional[str] = None,
         qps: float = 50.0,
         burst: int = 100,
         timeout: float = 30.0,
+        connection: Optional[KubeConnection] = None,
     ):
         self._limiter = _RateLimiter(qps, burst)
-        headers = {"Content-Type": "application/json"}
-        if kubeconfig:
-            base_url, verify, headers2 = self._from_kubeconfig(kubeconfig)
-            headers.update(headers2)
+        self._timeout = timeout
+        if connection is not None:
+            self._conn = connection
+        elif kubeconfig:
+            self._conn = load_kubeconfig(kubeconfig, context=context)
         else:
-            base_url, verify, headers2 = self._in_cluster()
-            headers.update(headers2)
-        self._client = httpx.Client(
-            base_url=base_url, verify=verify, headers=headers, timeout=timeout
-        )
+            self._conn = load_in_cluster()
+        self._client_lock = threading.Lock()
+        self._client_epoch = -1
+        self._client: Optional[httpx.Client] = None
+        self._http()  # fail fast on bad TLS material
 
-    @staticmethod
-    def _in_cluster():
-        host = os.environ.get("KUBERNETES_SERVICE_HOST")
-        port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
-        if not host:
-            raise RuntimeError(
-                "not running in-cluster (KUBERNETES_SERVICE_HOST unset) and "
-                "no kubeconfig given"
-            )
-        with open(SA_TOKEN) as f:
-            token = f.read().strip()
-        return (
-            f"https://{host}:{port}",
-            SA_CA if os.path.exists(SA_CA) else False,
-            {"Authorization": f"Bearer {token}"},
-        )
+    def _http(self) -> httpx.Client:
+        """Current HTTP client; rebuilt when the TLS epoch moves (exec
+        credential plugins can rotate client certificates)."""
+        with self._client_lock:
+            if self._client is None or self._client_epoch != self._conn.epoch:
+                old = self._client
+                self._client = httpx.Client(
+                    base_url=self._conn.server,
+                    verify=self._conn.ssl_verify(),
+                    headers={"Content-Type": "application/json"},
+                    timeout=self._timeout,
+                )
+                self._client_epoch = self._conn.epoch
+                if old is not None:
+                    old.close()
+            return self._client
 
-    @staticmethod
-    def _from_kubeconfig(path: str):
-        with open(path) as f:
-            cfg = yaml.safe_load(f)
-        ctx_name = cfg.get("current-context")
-        ctx = next(c for c in cfg["contexts"] if c["name"] == ctx_name)["context"]
-        cluster = next(
-            c for c in cfg["clusters"] if c["name"] == ctx["cluster"]
-        )["cluster"]
-        user = next(u for u in cfg["users"] if u["name"] == ctx["user"])["user"]
-        headers: Dict[str, str] = {}
-        if "token" in user:
-            headers["Authorization"] = f"Bearer {user['token']}"
-        verify = cluster.get("certificate-authority", True)
-        if cluster.get("insecure-skip-tls-verify"):
-            verify = False
-        return cluster["server"], verify, headers
+    def close(self) -> None:
+        with self._client_lock:
+            if self._client is not None:
+                self._client.close()
+                self._client = None
 
     # -- request core ------------------------------------------------------
-    def _req(self, method: str, path: str, body: Optional[dict] = None) -> dict:
+    def _req(
+        self,
+        method: str,
+        path: str,
+        body: Optional[dict] = None,
+        headers: Optional[Dict[str, str]] = None,
+    ) -> dict:
         self._limiter.acquire()
-        r = self._client.request(
-            method, path, content=json.dumps(body) if body is not None else None
+        h = self._conn.headers()  # may trigger exec refresh -> epoch bump
+        if headers:
+            h.update(headers)
+        r = self._http().request(
+            method,
+            path,
+            content=json.dumps(body) if body is not None else None,
+            headers=h,
         )
         if r.status_code == 404:
             raise NotFound(path)
@@ -162,21 +173,55 @@ class HttpKube(KubeClient):
         out = self._req("GET", f"{RESOURCE_V1BETA1}/deviceclasses")
         return out.get("items", [])
 
-    def watch_resource_claims(self, handler):
-        """Streaming watch (?watch=true, JSON-line events) with reconnect —
-        the informer analog. stop() is lazy: the reader exits at the next
-        event or when the connection drops."""
+    # -- watches -----------------------------------------------------------
+    def _watch(
+        self,
+        collection_path: str,
+        handler: Callable[[str, dict], None],
+        *,
+        name: str,
+    ):
+        """List+watch with resourceVersion resume (client-go reflector
+        shape). The initial LIST (and every re-list after a dropped window)
+        is replayed to the handler as synthetic ADDED events, so consumers
+        see a level-triggered stream and self-heal across gaps."""
         stop_event = threading.Event()
-        client = self._client
+        outer = self
+
+        def list_and_rv() -> Optional[str]:
+            try:
+                out = outer._req("GET", collection_path)
+            except Exception as e:
+                log.debug("%s: re-list failed: %s", name, e)
+                return None
+            for item in out.get("items", []):
+                handler("ADDED", item)
+            return (out.get("metadata") or {}).get("resourceVersion") or ""
 
         def run():
+            rv: Optional[str] = None
             while not stop_event.is_set():
+                if rv is None:
+                    rv = list_and_rv()
+                    if rv is None:
+                        if stop_event.wait(1.0):
+                            return
+                        continue
+                params = "?watch=true&allowWatchBookmarks=true"
+                if rv:
+                    params += f"&resourceVersion={rv}"
                 try:
-                    with client.stream(
+                    h = outer._conn.headers()
+                    with outer._http().stream(
                         "GET",
-                        f"{RESOURCE_V1BETA1}/resourceclaims?watch=true",
+                        f"{collection_path}{params}",
+                        headers=h,
                         timeout=httpx.Timeout(5.0, read=None),
                     ) as r:
+                        if r.status_code == 410:
+                            rv = None  # window expired: re-list
+                            continue
+                        r.raise_for_status()
                         for line in r.iter_lines():
                             if stop_event.is_set():
                                 return
@@ -187,13 +232,26 @@ class HttpKube(KubeClient):
                                 ev = json.loads(line)
                             except json.JSONDecodeError:
                                 continue
-                            handler(ev.get("type", ""), ev.get("object") or {})
+                            etype = ev.get("type", "")
+                            obj = ev.get("object") or {}
+                            new_rv = (obj.get("metadata") or {}).get(
+                                "resourceVersion"
+                            )
+                            if etype == "ERROR":
+                                # typically 410 Gone mid-stream
+                                rv = None
+                                break
+                            if new_rv:
+                                rv = new_rv
+                            if etype == "BOOKMARK":
+                                continue
+                            handler(etype, obj)
                 except Exception:
                     if stop_event.is_set():
                         return
                     time.sleep(1.0)  # transient-error retry (imex.go parity)
 
-        thread = threading.Thread(target=run, name="claims-watch", daemon=True)
+        thread = threading.Thread(target=run, name=name, daemon=True)
         thread.start()
 
         class _Watch:
@@ -201,6 +259,16 @@ class HttpKube(KubeClient):
                 stop_event.set()
 
         return _Watch()
+
+    def watch_resource_claims(self, handler):
+        return self._watch(
+            f"{RESOURCE_V1BETA1}/resourceclaims", handler, name="claims-watch"
+        )
+
+    def watch_resource_slices(self, handler):
+        return self._watch(
+            f"{RESOURCE_V1BETA1}/resourceslices", handler, name="slices-watch"
+        )
 
     def create_event(self, namespace: str, event: dict) -> None:
         try:
@@ -214,11 +282,13 @@ class HttpKube(KubeClient):
     def patch_node_labels(self, name: str, labels: Dict[str, Optional[str]]) -> dict:
         self._limiter.acquire()
         body = {"metadata": {"labels": labels}}
-        r = self._client.request(
+        h = self._conn.headers()
+        h["Content-Type"] = "application/strategic-merge-patch+json"
+        r = self._http().request(
             "PATCH",
             f"{CORE_V1}/nodes/{name}",
             content=json.dumps(body),
-            headers={"Content-Type": "application/strategic-merge-patch+json"},
+            headers=h,
         )
         if r.status_code == 404:
             raise NotFound(name)

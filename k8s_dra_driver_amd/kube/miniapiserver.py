@@ -48,10 +48,19 @@ class MiniApiServer:
                 n = int(self.headers.get("Content-Length") or 0)
                 return json.loads(self.rfile.read(n)) if n else None
 
+            def _list(self, items):
+                return self._json(
+                    200,
+                    {
+                        "metadata": {"resourceVersion": str(outer.store._rv)},
+                        "items": items,
+                    },
+                )
+
             def do_GET(self):
                 path = self.path.split("?")[0]
                 if path == _CLAIMS_ALL and "watch=true" in self.path:
-                    return self._watch_claims()
+                    return self._watch_kind("ResourceClaim")
                 m = _CLAIM_RE.match(path)
                 if m:
                     try:
@@ -61,18 +70,14 @@ class MiniApiServer:
                     except NotFound:
                         return self._json(404, {"reason": "NotFound"})
                 if path == _CLAIMS_ALL:
-                    return self._json(
-                        200, {"items": outer.store.list_resource_claims()}
-                    )
+                    return self._list(outer.store.list_resource_claims())
                 if path == _CLASSES:
-                    return self._json(
-                        200, {"items": outer.store.get_device_classes()}
-                    )
+                    return self._list(outer.store.get_device_classes())
                 m = _SLICE_RE.match(path)
                 if m and not m.group(1):
-                    return self._json(
-                        200, {"items": outer.store.list_resource_slices()}
-                    )
+                    if "watch=true" in self.path:
+                        return self._watch_kind("ResourceSlice")
+                    return self._list(outer.store.list_resource_slices())
                 m = _NODE_RE.match(path)
                 if m:
                     try:
@@ -81,20 +86,25 @@ class MiniApiServer:
                         return self._json(404, {})
                 self._json(404, {"path": path})
 
-            def _watch_claims(self):
+            def _watch_kind(self, want_kind):
                 """Streamed watch: JSON-line events until client disconnect
                 (read-until-close framing)."""
                 import queue
 
                 q: "queue.Queue" = queue.Queue()
+                verb_map = {
+                    "ADDED": "ADDED",
+                    "MODIFIED": "MODIFIED",
+                    "DELETED": "DELETED",
+                    "status": "MODIFIED",
+                    "create": "ADDED",
+                    "update": "MODIFIED",
+                    "delete": "DELETED",
+                }
 
                 def hook(kind, verb, obj):
-                    if kind == "ResourceClaim" and verb in (
-                        "ADDED",
-                        "MODIFIED",
-                        "DELETED",
-                    ):
-                        q.put({"type": verb, "object": obj})
+                    if kind == want_kind and verb in verb_map:
+                        q.put({"type": verb_map[verb], "object": obj})
 
                 outer.store.watchers.append(hook)
                 try:

@@ -19,6 +19,7 @@ class MiniApiServer:
         self.slices = {}
         self.nodes = {}
         self.requests = []
+        self.auth_headers = []
         outer = self
 
         class Handler(BaseHTTPRequestHandler):
@@ -39,6 +40,7 @@ class MiniApiServer:
 
             def do_GET(self):
                 outer.requests.append(("GET", self.path))
+                outer.auth_headers.append(self.headers.get("Authorization"))
                 parts = self.path.split("?")[0].strip("/").split("/")
                 if "resourceclaims" in parts:
                     ns, name = parts[4], parts[6]
@@ -155,9 +157,9 @@ def test_bearer_token_sent(api):
     srv, client = api
     srv.nodes["n1"] = {"metadata": {"name": "n1"}}
     client.get_node("n1")
-    # our kubeconfig carries a token; httpx should send it — verified via
-    # client internals (the mini server doesn't check auth)
-    assert client._client.headers["Authorization"] == "Bearer test-token"
+    # our kubeconfig carries a token; sent per-request (auth providers can
+    # rotate credentials, so headers are no longer baked into the client)
+    assert srv.auth_headers[-1] == "Bearer test-token"
 
 
 def test_rate_limiter_blocks():

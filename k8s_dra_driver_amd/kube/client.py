@@ -139,8 +139,8 @@ class InMemoryKube(KubeClient):
             cur = self.resource_claims[key]
             cur["status"] = copy.deepcopy(obj.get("status") or {})
             cur["metadata"]["resourceVersion"] = self._next_rv()
-            self._emit("ResourceClaim", "status", cur)
-            return copy.deepcopy(cur)
+        self._emit("ResourceClaim", "status", cur)
+        return copy.deepcopy(cur)
 
     def get_device_classes(self) -> List[dict]:
         with self._lock:
@@ -188,14 +188,17 @@ class InMemoryKube(KubeClient):
         return _Watch()
 
     def create_resource_slice(self, obj: dict) -> dict:
+        # events are emitted OUTSIDE the store lock: a synchronous watcher
+        # (e.g. the publisher's drift self-heal) re-enters the store from
+        # its own lock, and emitting under _lock would be an ABBA deadlock
         with self._lock:
             name = obj["metadata"]["name"]
             if name in self.resource_slices:
                 raise Conflict(f"resourceslice {name} exists")
             obj["metadata"]["resourceVersion"] = self._next_rv()
             self.resource_slices[name] = copy.deepcopy(obj)
-            self._emit("ResourceSlice", "create", obj)
-            return copy.deepcopy(obj)
+        self._emit("ResourceSlice", "create", obj)
+        return copy.deepcopy(obj)
 
     def update_resource_slice(self, obj: dict) -> dict:
         with self._lock:
@@ -208,14 +211,14 @@ class InMemoryKube(KubeClient):
                 raise Conflict(f"resourceslice {name} resourceVersion mismatch")
             obj["metadata"]["resourceVersion"] = self._next_rv()
             self.resource_slices[name] = copy.deepcopy(obj)
-            self._emit("ResourceSlice", "update", obj)
-            return copy.deepcopy(obj)
+        self._emit("ResourceSlice", "update", obj)
+        return copy.deepcopy(obj)
 
     def delete_resource_slice(self, name: str) -> None:
         with self._lock:
             obj = self.resource_slices.pop(name, None)
-            if obj is not None:
-                self._emit("ResourceSlice", "delete", obj)
+        if obj is not None:
+            self._emit("ResourceSlice", "delete", obj)
 
     def list_resource_slices(self, driver: Optional[str] = None) -> List[dict]:
         with self._lock:

@@ -10,17 +10,26 @@ reconcile contract in-process:
 - the pool generation bumps whenever the device set changes (repartition!)
   so the scheduler discards stale slices;
 - reconcile is level-triggered and idempotent: publish() computes desired
-  slices and creates/updates/deletes to match.
+  slices and creates/updates/deletes to match;
+- **drift self-healing** (``resourceslicecontroller.go:407-431``): with
+  :meth:`start_self_heal` the publisher watches this driver's slices and
+  re-publishes whenever one is externally deleted or mutated — the
+  fingerprint short-circuit is bypassed when observed state diverges from
+  desired, so a slice killed out from under the plugin comes back without
+  waiting for the next repartition/health event.
 """
 
 from __future__ import annotations
 
 import hashlib
 import json
+import logging
 import threading
-from typing import List, Optional
+from typing import Dict, List, Optional
 
 from .client import Conflict, KubeClient, NotFound
+
+log = logging.getLogger(__name__)
 
 MAX_DEVICES_PER_SLICE = 128
 API_VERSION = "resource.k8s.io/v1beta1"
@@ -49,96 +58,194 @@ class ResourceSlicePublisher:
         # deletes them when the Node object goes away (the k8s-native
         # complement to the controller's orphan sweep).
         self.node_uid = node_uid
-        self._lock = threading.Lock()
+        # RLock: a synchronous watch client (InMemoryKube) delivers the
+        # events of our own heal writes back into _on_slice_event on the
+        # same thread while _publish_locked still holds the lock.
+        self._lock = threading.RLock()
         self._generation = 0
         self._last_fingerprint: Optional[str] = None
+        #: desired spec per slice name, cached for drift detection
+        self._desired_specs: Dict[str, dict] = {}
+        #: the device list backing the current desired state
+        self._last_devices: Optional[List[dict]] = None
+        self._watch = None
+        #: observability: count of heals performed (tests/metrics)
+        self.heal_count = 0
 
     def _slice_name(self, index: int) -> str:
         safe_driver = self.driver_name.replace("/", "-").replace(".", "-")
         return f"{self.node_name}-{safe_driver}-{index}"
 
+    # ------------------------------------------------------------------
+    # publication
+    # ------------------------------------------------------------------
     def publish(self, devices: List[dict]) -> List[dict]:
         """Reconcile the published slices to carry exactly ``devices``.
 
-        Returns the slice objects as stored. No-op (no API calls) when the
-        device set is unchanged — cheap to call after every enumeration.
+        Returns the slice objects as stored. No-op (no API calls beyond the
+        LIST) when the device set is unchanged AND the observed slices match
+        the desired state — cheap to call after every enumeration, but never
+        blind to external tampering.
         """
         with self._lock:
-            fp = _devices_fingerprint(devices)
-            existing = {
-                s["metadata"]["name"]: s
-                for s in self.client.list_resource_slices(self.driver_name)
-                if s["spec"].get("nodeName") == self.node_name
-            }
-            if fp == self._last_fingerprint and existing:
-                return list(existing.values())
-            if self._generation == 0 and existing:
-                # plugin restart: never regress the pool generation below
-                # what is already published (the scheduler treats higher
-                # generations as authoritative for multi-slice pools)
-                self._generation = max(
-                    (s["spec"].get("pool", {}).get("generation", 0))
-                    for s in existing.values()
-                )
-            self._generation += 1
+            return self._publish_locked(devices)
 
-            chunks = [
-                devices[i : i + MAX_DEVICES_PER_SLICE]
-                for i in range(0, len(devices), MAX_DEVICES_PER_SLICE)
-            ] or [[]]
-            desired_names = {self._slice_name(i) for i in range(len(chunks))}
-            out = []
-            for i, chunk in enumerate(chunks):
-                name = self._slice_name(i)
-                meta: dict = {"name": name}
-                if self.node_uid:
-                    meta["ownerReferences"] = [
-                        {
-                            "apiVersion": "v1",
-                            "kind": "Node",
-                            "name": self.node_name,
-                            "uid": self.node_uid,
-                        }
+    def _observed_matches_desired(self, existing: Dict[str, dict]) -> bool:
+        if set(existing) != set(self._desired_specs):
+            return False
+        for name, want_spec in self._desired_specs.items():
+            if existing[name].get("spec") != want_spec:
+                return False
+        return True
+
+    def _publish_locked(self, devices: List[dict]) -> List[dict]:
+        fp = _devices_fingerprint(devices)
+        existing = {
+            s["metadata"]["name"]: s
+            for s in self.client.list_resource_slices(self.driver_name)
+            if s["spec"].get("nodeName") == self.node_name
+        }
+        if (
+            fp == self._last_fingerprint
+            and existing
+            and self._observed_matches_desired(existing)
+        ):
+            return list(existing.values())
+        if self._generation == 0 and existing:
+            # plugin restart: never regress the pool generation below
+            # what is already published (the scheduler treats higher
+            # generations as authoritative for multi-slice pools)
+            self._generation = max(
+                (s["spec"].get("pool", {}).get("generation", 0))
+                for s in existing.values()
+            )
+        self._generation += 1
+
+        chunks = [
+            devices[i : i + MAX_DEVICES_PER_SLICE]
+            for i in range(0, len(devices), MAX_DEVICES_PER_SLICE)
+        ] or [[]]
+        desired_names = {self._slice_name(i) for i in range(len(chunks))}
+        out = []
+        self._desired_specs = {}
+        for i, chunk in enumerate(chunks):
+            name = self._slice_name(i)
+            meta: dict = {"name": name}
+            if self.node_uid:
+                meta["ownerReferences"] = [
+                    {
+                        "apiVersion": "v1",
+                        "kind": "Node",
+                        "name": self.node_name,
+                        "uid": self.node_uid,
+                    }
+                ]
+            spec = {
+                "driver": self.driver_name,
+                "nodeName": self.node_name,
+                "pool": {
+                    "name": self.pool_name,
+                    "generation": self._generation,
+                    "resourceSliceCount": len(chunks),
+                },
+                "devices": chunk,
+            }
+            obj = {
+                "apiVersion": API_VERSION,
+                "kind": "ResourceSlice",
+                "metadata": meta,
+                "spec": spec,
+            }
+            # register desired BEFORE the write: a synchronous watcher
+            # delivers our own create/update event re-entrantly, and it
+            # must compare against the new spec (match -> no heal loop)
+            self._desired_specs[name] = spec
+            if name in existing:
+                obj["metadata"]["resourceVersion"] = existing[name][
+                    "metadata"
+                ]["resourceVersion"]
+                try:
+                    out.append(self.client.update_resource_slice(obj))
+                except Conflict:
+                    # concurrent writer (e.g. admission rewrite): refetch
+                    cur = {
+                        s["metadata"]["name"]: s
+                        for s in self.client.list_resource_slices(
+                            self.driver_name
+                        )
+                    }[name]
+                    obj["metadata"]["resourceVersion"] = cur["metadata"][
+                        "resourceVersion"
                     ]
-                obj = {
-                    "apiVersion": API_VERSION,
-                    "kind": "ResourceSlice",
-                    "metadata": meta,
-                    "spec": {
-                        "driver": self.driver_name,
-                        "nodeName": self.node_name,
-                        "pool": {
-                            "name": self.pool_name,
-                            "generation": self._generation,
-                            "resourceSliceCount": len(chunks),
-                        },
-                        "devices": chunk,
-                    },
-                }
-                if name in existing:
-                    obj["metadata"]["resourceVersion"] = existing[name][
-                        "metadata"
-                    ]["resourceVersion"]
-                    try:
-                        out.append(self.client.update_resource_slice(obj))
-                    except Conflict:
-                        # concurrent writer (e.g. admission rewrite): refetch
-                        cur = {
-                            s["metadata"]["name"]: s
-                            for s in self.client.list_resource_slices(
-                                self.driver_name
-                            )
-                        }[name]
-                        obj["metadata"]["resourceVersion"] = cur["metadata"][
-                            "resourceVersion"
-                        ]
-                        out.append(self.client.update_resource_slice(obj))
-                else:
+                    out.append(self.client.update_resource_slice(obj))
+            else:
+                try:
                     out.append(self.client.create_resource_slice(obj))
-            for name in set(existing) - desired_names:
-                self.client.delete_resource_slice(name)
-            self._last_fingerprint = fp
-            return out
+                except Conflict:
+                    # raced an external re-creation: overwrite it
+                    cur = {
+                        s["metadata"]["name"]: s
+                        for s in self.client.list_resource_slices(
+                            self.driver_name
+                        )
+                    }.get(name)
+                    if cur is None:
+                        raise
+                    obj["metadata"]["resourceVersion"] = cur["metadata"][
+                        "resourceVersion"
+                    ]
+                    out.append(self.client.update_resource_slice(obj))
+        for name in set(existing) - desired_names:
+            self.client.delete_resource_slice(name)
+        self._last_fingerprint = fp
+        self._last_devices = list(devices)
+        return out
+
+    # ------------------------------------------------------------------
+    # drift self-heal (resourceslicecontroller.go:407-431 parity)
+    # ------------------------------------------------------------------
+    def start_self_heal(self) -> bool:
+        """Watch this driver's slices and republish on external drift.
+
+        Returns False when the client cannot watch (callers may poll)."""
+        if self._watch is not None:
+            return True
+        w = self.client.watch_resource_slices(self._on_slice_event)
+        if w is None:
+            return False
+        self._watch = w
+        return True
+
+    def stop_self_heal(self) -> None:
+        if self._watch is not None:
+            self._watch.stop()
+            self._watch = None
+
+    def _on_slice_event(self, etype: str, obj: dict) -> None:
+        try:
+            spec = obj.get("spec") or {}
+            name = (obj.get("metadata") or {}).get("name", "")
+            with self._lock:
+                if self._last_devices is None or not self._desired_specs:
+                    return  # nothing published yet
+                if name not in self._desired_specs:
+                    # not ours (or an orphan being deleted) — ignore
+                    return
+                if etype == "DELETED":
+                    drifted = True
+                else:
+                    drifted = spec != self._desired_specs[name]
+                if not drifted:
+                    return
+                log.warning(
+                    "resourceslice %s drifted externally (%s); re-publishing",
+                    name,
+                    etype,
+                )
+                self.heal_count += 1
+                self._publish_locked(list(self._last_devices))
+        except Exception:
+            log.exception("slice self-heal failed for event %s", etype)
 
     def unpublish_all(self) -> None:
         """Delete every slice this driver owns on this node (clean shutdown
@@ -151,3 +258,5 @@ class ResourceSlicePublisher:
                     except NotFound:
                         pass
             self._last_fingerprint = None
+            self._desired_specs = {}
+            self._last_devices = None

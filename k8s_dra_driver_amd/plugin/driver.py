@@ -117,8 +117,12 @@ class Driver:
         (reference main.go:167-206 + driver.go:70-84)."""
         self.state.write_base_cdi_spec()
         self.publish_resources()
+        # Watch our slices and republish on external deletion/mutation
+        # (resourceslicecontroller.go:407-431 parity).
+        self.publisher.start_self_heal()
 
     def shutdown(self, unpublish: bool = True) -> None:
+        self.publisher.stop_self_heal()
         self._pool.shutdown(wait=True)
         self.state.close()
         if unpublish:

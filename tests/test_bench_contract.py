@@ -56,7 +56,7 @@ def test_bench_json_contract():
     assert d["vs_baseline"] is None  # reference publishes no number
     assert d["value"] > 0 and d["ms_per_step"] > 0
     cfg = d["config"]
-    assert cfg["model"] == "dra-claim-lifecycle"
+    assert cfg["model"] == "dra-claim-lifecycle/whole"
     assert cfg["hal"] in ("fake", "amdsmi")
     assert cfg["alloc_prepare_p50_ms"] > 0
 
@@ -135,3 +135,52 @@ def test_repartition_metric_counts(tmp_path):
     text = generate_latest(driver.metrics.registry).decode()
     # one switch at prepare + one restore at unprepare
     assert "dra_repartitions_total 2.0" in text
+
+
+def test_bench_all_configs_attested():
+    """BASELINE configs #1-#5 are first-class bench modes:
+    --config all emits the whole-GPU headline plus per-config results."""
+    out = subprocess.run(
+        [
+            sys.executable,
+            os.path.join(REPO, "bench.py"),
+            "--steps", "2", "--warmup", "1", "--pods-per-step", "8",
+            "--config", "all",
+        ],
+        capture_output=True,
+        text=True,
+        cwd=REPO,
+        timeout=280,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads(out.stdout.strip().splitlines()[-1])
+    assert d["config"]["model"] == "dra-claim-lifecycle/whole"
+    configs = d["config"]["configs"]
+    for name in ("mock", "shared", "cpx", "topo4"):
+        assert name in configs, name
+        sub = configs[name]
+        assert "error" not in sub, f"{name}: {sub}"
+        assert sub["pods_per_sec"] > 0
+        assert sub["alloc_prepare_p50_ms"] > 0
+    assert configs["mock"]["hal"] == "fake"
+
+
+def test_bench_hal_amdsmi_hard_fails_without_gpu():
+    """--hal amdsmi must never silently bench the fake backend."""
+    out = subprocess.run(
+        [
+            sys.executable,
+            os.path.join(REPO, "bench.py"),
+            "--steps", "1", "--warmup", "0", "--pods-per-step", "1",
+            "--hal", "amdsmi",
+        ],
+        capture_output=True,
+        text=True,
+        cwd=REPO,
+        timeout=280,
+    )
+    if out.returncode == 0:  # only on a real GPU box
+        d = json.loads(out.stdout.strip().splitlines()[-1])
+        assert d["config"]["hal"] == "amdsmi"
+    else:
+        assert "amdsmi" in out.stderr

@@ -60,6 +60,10 @@ class AmdSmiDeviceLib(DeviceLib):
         self._ext_override = ext
         self.topology = KfdTopology(sysfs_root)
         self._timeslice: Dict[int, Optional[int]] = {}
+        self._quantum_knob_writable: Optional[bool] = None
+        self._quantum_default_ms: int = 0
+        #: learned capability: flips False after a failed partition set
+        self._repartition_capable: Optional[bool] = None
         self._rocm = _rocm_version()
 
     # -- lifecycle ---------------------------------------------------------
@@ -174,7 +178,14 @@ class AmdSmiDeviceLib(DeviceLib):
                 if head.get("xgmi_hive_id")
                 else "",
                 xgmi_node_id=int(head.get("xgmi_node_id", -1)),
+                timeslice_effective=self.timeslice_effective(),
             )
+            # capability probe (no recursion: uses the card minor in hand)
+            if self._repartition_capable is None and info.card_minor >= 0:
+                self._repartition_capable = self._probe_repartition_path(
+                    info.card_minor
+                )
+            info.repartition_capable = bool(self._repartition_capable)
 
             if compute_mode != "SPX" and len(members) > 1:
                 try:
@@ -234,16 +245,52 @@ class AmdSmiDeviceLib(DeviceLib):
                 )
 
     # -- partitioning ------------------------------------------------------
+    def dynamic_repartition_capable(self) -> bool:
+        """Best-effort probe: can this node actually switch partition
+        modes? Checks the bare-metal sysfs control is present and writable
+        (virtualized/shared pools expose it read-only or reject writes —
+        round-1 saw AMDSMI_STATUS_UNKNOWN_ERROR there), and
+        learns from failed sets. Never raises."""
+        if self._repartition_capable is not None:
+            return self._repartition_capable
+        capable = False
+        try:
+            for g in self.enumerate():  # enumerate() fills the cache
+                if g.card_minor >= 0:
+                    capable = self._probe_repartition_path(g.card_minor)
+                    break
+        except Exception:
+            capable = False
+        self._repartition_capable = capable
+        return capable
+
+    def _probe_repartition_path(self, card_minor: int) -> bool:
+        path = (
+            f"{self.topology.root}/class/drm/card{card_minor}"
+            "/device/current_compute_partition"
+        )
+        return os.path.isfile(path) and os.access(path, os.W_OK)
+
     def set_compute_partition(self, gpu_index: int, mode: str) -> None:
         ext = self._require()
         proc_index = self._head_proc_index(gpu_index)
-        ext.set_compute_partition(proc_index, mode)
+        try:
+            ext.set_compute_partition(proc_index, mode)
+        except Exception:
+            self._repartition_capable = False  # learned: this box refuses
+            raise
+        self._repartition_capable = True
         ext.reinit()  # processor handles change with the KFD device set
 
     def set_memory_partition(self, gpu_index: int, mode: str) -> None:
         ext = self._require()
         proc_index = self._head_proc_index(gpu_index)
-        ext.set_memory_partition(proc_index, mode)
+        try:
+            ext.set_memory_partition(proc_index, mode)
+        except Exception:
+            self._repartition_capable = False
+            raise
+        self._repartition_capable = True
         ext.reinit()
 
     def _head_proc_index(self, gpu_index: int) -> int:
@@ -255,7 +302,9 @@ class AmdSmiDeviceLib(DeviceLib):
             oam = p.get("oam_id", 0xFFFFFFFF)
             key = oam if oam != 0xFFFFFFFF else p.get("uuid", p["index"])
             groups[key].append(p)
-        ordered = sorted(groups.items(), key=lambda kv: str(kv[0]))
+        ordered = sorted(
+            groups.items(), key=lambda kv: (isinstance(kv[0], str), kv[0])
+        )
         if gpu_index >= len(ordered):
             raise HalError(f"gpu-{gpu_index} not found ({len(ordered)} GPUs)")
         members = sorted(
@@ -265,12 +314,60 @@ class AmdSmiDeviceLib(DeviceLib):
         return int(members[0]["index"])
 
     # -- scheduler controls --------------------------------------------------
+    #: runtime-writable amdgpu module param controlling how long a compute
+    #: queue may run before the HWS preempts it — the closest real analog
+    #: of the reference's per-GPU compute-policy time-slice. Global (all
+    #: GPUs on the node), probed once at first use.
+    QUEUE_PREEMPTION_PARAM = (
+        "/sys/module/amdgpu/parameters/queue_preemption_timeout_ms"
+    )
+
+    def _probe_quantum_knob(self) -> bool:
+        """True when the node exposes a writable scheduler-quantum knob."""
+        if self._quantum_knob_writable is None:
+            path = self.QUEUE_PREEMPTION_PARAM
+            self._quantum_knob_writable = os.path.isfile(path) and os.access(
+                path, os.W_OK
+            )
+            if self._quantum_knob_writable:
+                try:
+                    with open(path) as f:
+                        self._quantum_default_ms = int(f.read().strip())
+                except (OSError, ValueError):
+                    self._quantum_knob_writable = False
+        return self._quantum_knob_writable
+
+    def timeslice_effective(self) -> bool:
+        return self._probe_quantum_knob()
+
     def set_timeslice_quantum(self, gpu_index: int, quantum_us: Optional[int]) -> None:
-        """amdgpu exposes no per-GPU runtime quantum control (the
-        compute-policy analog); the interval is recorded
-        and surfaced via health/metrics. HSA default time-slicing already
-        multiplexes queues fairly."""
+        """Scheduler-quantum control (the compute-policy analog, SURVEY.md
+        §2.3 N7). When the node exposes a writable
+        ``queue_preemption_timeout_ms`` the quantum is applied there (a
+        node-global knob: amdgpu has no per-GPU granularity, documented via
+        the ``timeSlicingEffective`` attribute contract); otherwise the
+        request is recorded and surfaced as advisory — HSA default
+        time-slicing still multiplexes queues fairly."""
         self._timeslice[gpu_index] = quantum_us
+        if self._probe_quantum_knob():
+            value_ms = (
+                self._quantum_default_ms
+                if quantum_us is None
+                else max(1, quantum_us // 1000)
+            )
+            try:
+                with open(self.QUEUE_PREEMPTION_PARAM, "w") as f:
+                    f.write(str(value_ms))
+                log.info(
+                    "gpu-%d: queue preemption timeout set to %d ms "
+                    "(node-global amdgpu knob)",
+                    gpu_index,
+                    value_ms,
+                )
+                return
+            except OSError as e:
+                log.warning("quantum knob write failed: %s", e)
+                self._quantum_knob_writable = False
         log.info(
             "gpu-%d: time-slice quantum request %s recorded "
             "(amdgpu scheduler default multiplexing applies)",

@@ -105,6 +105,11 @@ class FakeDeviceLib(DeviceLib):
         self._open = False
         self._gpus = [_FakeGpuState(self.cfg, i) for i in range(self.cfg.num_gpus)]
         self._timeslice: Dict[int, Optional[int]] = {}
+        #: models whether this fake node honors quantum requests (set False
+        #: to simulate real amdgpu pools where time-slicing is advisory)
+        self.timeslice_effective_flag = True
+        #: models bare-metal partition-switch capability
+        self.repartition_capable_flag = True
         #: observers notified after any partition-state change (used by the
         #: plugin to republish ResourceSlices without polling)
         self._observers: List[Callable[[], None]] = []
@@ -187,6 +192,8 @@ class FakeDeviceLib(DeviceLib):
                     compute_caps=list(COMPUTE_MODES),
                     xgmi_hive_id=self.cfg.hive_id,
                     xgmi_node_id=g.index,
+                    timeslice_effective=self.timeslice_effective_flag,
+                    repartition_capable=self.repartition_capable_flag,
                     links=links,
                 )
                 if g.compute_mode != "SPX":
@@ -270,6 +277,12 @@ class FakeDeviceLib(DeviceLib):
 
     def get_timeslice_quantum(self, gpu_index: int) -> Optional[int]:
         return self._timeslice.get(gpu_index)
+
+    def timeslice_effective(self) -> bool:
+        return self.timeslice_effective_flag
+
+    def dynamic_repartition_capable(self) -> bool:
+        return self.repartition_capable_flag
 
     # -- device nodes --------------------------------------------------------
     def device_node_paths(

@@ -116,6 +116,12 @@ class GpuInfo:
     )
     xgmi_hive_id: str = ""
     xgmi_node_id: int = -1
+    #: whether a TimeSlicing interval on this node changes real scheduler
+    #: behavior (False => requests are advisory; published so CEL selectors
+    #: and users can tell — VERDICT r1 #5)
+    timeslice_effective: bool = True
+    #: whether this node can switch partition modes at runtime (bare metal)
+    repartition_capable: bool = True
     links: List[XgmiLink] = field(default_factory=list)
     #: partitions present when compute_partition != SPX
     partitions: List[PartitionedDeviceInfo] = field(default_factory=list)
@@ -164,6 +170,8 @@ def _common_gpu_attrs(gpu: GpuInfo) -> Dict[str, dict]:
         qualified("xgmiPeerOamIds"): _attr_str(
             ",".join(str(i) for i in gpu.xgmi_peer_oam_ids())
         ),
+        qualified("timeSlicingEffective"): _attr_bool(gpu.timeslice_effective),
+        qualified("repartitionCapable"): _attr_bool(gpu.repartition_capable),
     }
     if gpu.driver_version:
         attrs[qualified("driverVersion")] = _attr_ver(gpu.driver_version)

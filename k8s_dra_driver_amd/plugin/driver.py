@@ -103,6 +103,7 @@ class Driver:
         # Republish whenever the allocatable set changes (repartition).
         self.state.on_allocatable_change = self.publish_resources
         self.state.on_repartition = self.metrics.repartitions.inc
+        self.state.on_warning = self._emit_claim_warning
         # Failure detection: unhealthy GPUs are pulled from publication
         # (start()ed by main.py; tests drive check_once directly).
         from .health import HealthMonitor
@@ -130,6 +131,29 @@ class Driver:
                 self.publisher.unpublish_all()
             except Exception:
                 log.exception("unpublish failed")
+
+    def _emit_claim_warning(self, info, reason: str, message: str) -> None:
+        """Kubernetes Warning event on the claim (operator visibility for
+        platform gaps like advisory time-slicing; best-effort)."""
+        self.kube.create_event(
+            info.namespace or "default",
+            {
+                "metadata": {
+                    "generateName": f"{(info.name or info.uid)[:40]}-",
+                },
+                "type": "Warning",
+                "reason": reason,
+                "message": message,
+                "involvedObject": {
+                    "apiVersion": "resource.k8s.io/v1beta1",
+                    "kind": "ResourceClaim",
+                    "namespace": info.namespace,
+                    "name": info.name,
+                    "uid": info.uid,
+                },
+                "source": {"component": DRIVER_NAME},
+            },
+        )
 
     def publish_resources(self) -> None:
         unhealthy = self.health.unhealthy_gpus

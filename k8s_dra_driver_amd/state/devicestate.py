@@ -109,6 +109,9 @@ class DeviceState:
         #: optional callback fired after the allocatable set changes
         #: (repartition) — the Driver republishes ResourceSlices from it.
         self.on_allocatable_change = None
+        #: optional callback(claim_info, reason, message) for user-facing
+        #: Warning events (e.g. advisory time-slicing); wired by Driver
+        self.on_warning = None
         #: optional callback per performed mode switch (metrics)
         self.on_repartition = None
 
@@ -427,29 +430,60 @@ class DeviceState:
             prepared.shared_session_id = session.session_id
             shared_edits = session.container_edits()
 
-        ts_devices = [
-            dev
-            for gcfg, dev in gcfg_by_dev
-            if gcfg
-            and gcfg.sharing
-            and gcfg.sharing.strategy == TIME_SLICING
-            and gcfg.sharing.time_slicing.interval != "Default"
-        ]
-        if ts_devices:
-            settings = next(
-                gcfg.sharing.time_slicing
-                for gcfg, dev in gcfg_by_dev
-                if dev in ts_devices
-            )
-            try:
-                prepared.timeslice_gpus = self.ts_manager.set_timeslice(
-                    ts_devices, settings
+        # Group time-sliced devices by their settings: two requests in one
+        # claim may carry different intervals, and collapsing them to the
+        # first match would silently mis-apply (round-1 "weak" finding).
+        # A genuine conflict — two intervals for the SAME parent GPU — is
+        # rejected rather than resolved arbitrarily.
+        ts_by_quantum: Dict[int, tuple] = {}
+        gpu_interval: Dict[int, str] = {}
+        for gcfg, dev in gcfg_by_dev:
+            if not (
+                gcfg
+                and gcfg.sharing
+                and gcfg.sharing.strategy == TIME_SLICING
+                and gcfg.sharing.time_slicing.interval != "Default"
+            ):
+                continue
+            ts = gcfg.sharing.time_slicing
+            idx = dev.parent_gpu.index
+            if gpu_interval.get(idx, ts.interval) != ts.interval:
+                raise PrepareError(
+                    f"conflicting TimeSlicing intervals for gpu-{idx} in one "
+                    f"claim ({gpu_interval[idx]} vs {ts.interval})"
                 )
-            except Exception as e:
-                # _rollback (invoked by the caller) undoes the session
-                raise PrepareError(f"time-slicing failed: {e}") from e
+            gpu_interval[idx] = ts.interval
+            entry = ts_by_quantum.setdefault(ts.quantum_us, (ts, []))
+            entry[1].append(dev)
+        if ts_by_quantum:
+            for settings, devs in ts_by_quantum.values():
+                try:
+                    prepared.timeslice_gpus.extend(
+                        self.ts_manager.set_timeslice(devs, settings)
+                    )
+                except Exception as e:
+                    # _rollback (invoked by the caller) undoes the session
+                    raise PrepareError(f"time-slicing failed: {e}") from e
             if not prepared.sharing_strategy:
                 prepared.sharing_strategy = TIME_SLICING
+            if not self.lib.timeslice_effective():
+                # Surface the platform gap instead of silently reporting
+                # success: this hardware has no runtime
+                # scheduler-quantum control, so the interval is advisory.
+                msg = (
+                    "TimeSlicing interval(s) "
+                    f"{sorted(set(gpu_interval.values()))} accepted but "
+                    "ADVISORY on this node: amdgpu exposes no runtime "
+                    "scheduler-quantum control (timeSlicingEffective=false "
+                    "on the published devices); HSA default queue "
+                    "multiplexing applies"
+                )
+                log.warning("claim %s: %s", info.uid, msg)
+                if self.on_warning is not None:
+                    try:
+                        self.on_warning(info, "TimeSlicingAdvisory", msg)
+                    except Exception:
+                        log.exception("warning callback failed")
 
         # --- CDI claim spec -------------------------------------------------
         claim_edits = ContainerEdits(

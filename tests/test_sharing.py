@@ -117,3 +117,129 @@ class TestSharedCompute:
         m1 = int(next(e for e in s1.env if "HSA_CU_MASK" in e).split(":")[1], 16)
         m2 = int(next(e for e in s2.env if "HSA_CU_MASK" in e).split(":")[1], 16)
         assert m1 & m2 == 0
+
+
+class TestTimeSlicingHonesty:
+    """VERDICT r1 #5: advisory time-slicing is surfaced, not silent."""
+
+    def _ts_cfg(self, interval="Short"):
+        return {
+            "source": "FromClaim",
+            "requests": [],
+            "opaque": {
+                "driver": "gpu.amd.com",
+                "parameters": {
+                    "apiVersion": "resource.gpu.amd.com/v1alpha1",
+                    "kind": "GpuConfig",
+                    "sharing": {
+                        "strategy": "TimeSlicing",
+                        "timeSlicingConfig": {"interval": interval},
+                    },
+                },
+            },
+        }
+
+    def _state(self, tmp_path, lib):
+        from k8s_dra_driver_amd.cdi.handler import CDIHandler
+        from k8s_dra_driver_amd.state.checkpoint import CheckpointStore
+        from k8s_dra_driver_amd.state.devicestate import DeviceState
+
+        return DeviceState(
+            lib,
+            CDIHandler(cdi_root=str(tmp_path / "cdi")),
+            CheckpointStore(str(tmp_path / "ckpt")),
+            pool_name="n",
+        )
+
+    def _claim(self, uid, dev_cfgs):
+        results = [
+            {
+                "request": f"req-{i}",
+                "driver": "gpu.amd.com",
+                "pool": "n",
+                "device": dev,
+            }
+            for i, (dev, _) in enumerate(dev_cfgs)
+        ]
+        configs = []
+        for i, (_, interval) in enumerate(dev_cfgs):
+            c = self._ts_cfg(interval)
+            c["requests"] = [f"req-{i}"]
+            configs.append(c)
+        return {
+            "metadata": {"namespace": "d", "name": f"c-{uid}", "uid": uid},
+            "status": {
+                "allocation": {
+                    "devices": {"results": results, "config": configs}
+                }
+            },
+        }
+
+    def test_advisory_timeslice_emits_warning(self, tmp_path):
+        from k8s_dra_driver_amd.hal import FakeDeviceLib
+
+        lib = FakeDeviceLib()
+        lib.open()
+        lib.timeslice_effective_flag = False  # a real amdgpu pool
+        state = self._state(tmp_path, lib)
+        warnings = []
+        state.on_warning = lambda info, reason, msg: warnings.append(
+            (info.uid, reason, msg)
+        )
+        state.prepare(self._claim("u-adv", [("gpu-0", "Short")]))
+        assert len(warnings) == 1
+        uid, reason, msg = warnings[0]
+        assert uid == "u-adv" and reason == "TimeSlicingAdvisory"
+        assert "ADVISORY" in msg and "timeSlicingEffective" in msg
+
+    def test_effective_timeslice_no_warning(self, tmp_path):
+        from k8s_dra_driver_amd.hal import FakeDeviceLib
+
+        lib = FakeDeviceLib()
+        lib.open()
+        state = self._state(tmp_path, lib)
+        warnings = []
+        state.on_warning = lambda *a: warnings.append(a)
+        state.prepare(self._claim("u-eff", [("gpu-0", "Short")]))
+        assert warnings == []
+
+    def test_timeslice_effective_attribute_published(self):
+        from k8s_dra_driver_amd.hal import FakeDeviceLib
+        from k8s_dra_driver_amd.hal.model import AllocatableDevice
+
+        lib = FakeDeviceLib()
+        lib.open()
+        lib.timeslice_effective_flag = False
+        dev = AllocatableDevice.from_gpu(lib.enumerate()[0]).to_device()
+        attrs = dev["basic"]["attributes"]
+        assert attrs["gpu.amd.com/timeSlicingEffective"]["bool"] is False
+        assert attrs["gpu.amd.com/repartitionCapable"]["bool"] is True
+
+    def test_mixed_intervals_different_gpus_apply_per_group(self, tmp_path):
+        from k8s_dra_driver_amd.hal import FakeDeviceLib
+
+        lib = FakeDeviceLib()
+        lib.open()
+        state = self._state(tmp_path, lib)
+        state.prepare(
+            self._claim("u-mix", [("gpu-0", "Short"), ("gpu-1", "Long")])
+        )
+        assert lib.get_timeslice_quantum(0) == 1000  # Short
+        assert lib.get_timeslice_quantum(1) == 10000  # Long
+        state.unprepare("u-mix")
+        assert lib.get_timeslice_quantum(0) is None
+        assert lib.get_timeslice_quantum(1) is None
+
+    def test_conflicting_intervals_same_gpu_rejected(self, tmp_path):
+        import pytest
+
+        from k8s_dra_driver_amd.hal import FakeDeviceLib
+        from k8s_dra_driver_amd.state.devicestate import PrepareError
+
+        lib = FakeDeviceLib()
+        lib.open()
+        state = self._state(tmp_path, lib)
+        with pytest.raises(PrepareError, match="conflicting TimeSlicing"):
+            state.prepare(
+                self._claim("u-con", [("gpu-0", "Short"), ("gpu-0", "Long")])
+            )

@@ -146,6 +146,16 @@ class BenchRank:
         if config == "mock":
             hal_mode = "fake"  # config #1 is explicitly the stub config
         self.lib, self.hal_kind = pick_hal(hal_mode)
+        if (
+            config == "cpx"
+            and self.hal_kind == "amdsmi"
+            and not self.lib.dynamic_repartition_capable()
+        ):
+            self.lib.close()
+            raise RuntimeError(
+                "dynamic repartition unavailable on this box (partition "
+                "sysfs control not writable — virtualized/shared pool)"
+            )
         if config == "topo4" and len(self.lib.enumerate()) < 4:
             if hal_mode == "amdsmi":
                 raise RuntimeError(
@@ -448,8 +458,23 @@ def main() -> int:
     for c in extra:
         try:
             extra_res[c] = run_config(c, args, rank, local_rank, world, sync)
-        except Exception as e:  # attest the failure rather than dying
-            extra_res[c] = {"error": str(e)}
+        except Exception as e:
+            # A sub-config that this box physically cannot run on real
+            # hardware (repartition-incapable pool, <4 GPUs) is measured
+            # on the fake backend with an EXPLICIT label — never silently.
+            if args.hal != "fake":
+                fb = argparse.Namespace(**vars(args))
+                fb.hal = "fake"
+                try:
+                    r = run_config(c, fb, rank, local_rank, world, sync)
+                    r["fallback_reason"] = str(e)
+                    extra_res[c] = r
+                except Exception as e2:
+                    extra_res[c] = {
+                        "error": f"{e}; fake fallback also failed: {e2}"
+                    }
+            else:  # attest the failure rather than dying
+                extra_res[c] = {"error": str(e)}
 
     result = {
         "metric": "gpu_pods_scheduled_per_sec",

@@ -237,3 +237,31 @@ class TestKfdBackendOnHardware:
         assert k.cu_count == s.cu_count == 256
         # VRAM from mem_banks within 2% of amdsmi's number
         assert abs(k.vram_total_mib - s.vram_total_mib) < s.vram_total_mib * 0.02
+
+
+class TestRcclFabric:
+    """RCCL-over-xGMI readiness: the allreduce demo the
+    multi-GPU topology claims run, wired as a gpu test that auto-skips
+    below world-size 2 (the pool's boxes expose 1 GPU; an 8-GPU node
+    exercises the full ring)."""
+
+    def test_rccl_allreduce_all_visible_gpus(self):
+        if torch.cuda.device_count() < 2:
+            pytest.skip("needs >=2 visible GPUs for a ring all-reduce")
+        from k8s_dra_driver_amd.workload import run_allreduce
+
+        assert run_allreduce(64) == 0
+
+    def test_timeslice_knob_probe_runs(self, real_lib):
+        """The quantum-knob probe must never raise on real hardware, and
+        the published attribute must reflect it."""
+        eff = real_lib.timeslice_effective()
+        assert isinstance(eff, bool)
+        g0 = real_lib.enumerate()[0]
+        assert g0.timeslice_effective == eff
+
+    def test_repartition_capability_probe_runs(self, real_lib):
+        cap = real_lib.dynamic_repartition_capable()
+        assert isinstance(cap, bool)
+        g0 = real_lib.enumerate()[0]
+        assert g0.repartition_capable == cap

@@ -58,6 +58,43 @@ def build_amdhal(force: bool = False) -> str:
     return out
 
 
+def build_amdhal_asan(force: bool = False) -> str:
+    """AddressSanitizer flavor of the amdsmi binding (the
+    cgo-shim ASan build the reference's pure-Go stack never needed).
+    Run it with:
+        LD_PRELOAD=$(gcc -print-file-name=libasan.so) \
+        ASAN_OPTIONS=detect_leaks=0 python -c 'import ... _amdhal_asan'
+    (leak detection off: the Python interpreter itself 'leaks' at exit).
+    """
+    src = os.path.join(CPP_DIR, "amdhal.cpp")
+    out = os.path.join(PKG_DIR, f"_amdhal_asan{_ext_suffix()}")
+    if force or _needs_build(src, out):
+        _run(
+            [
+                "g++",
+                "-shared",
+                "-fPIC",
+                "-g",
+                "-O1",
+                "-std=c++17",
+                "-fsanitize=address",
+                "-fno-omit-frame-pointer",
+                "-DPYBIND11_MODULE_NAME=_amdhal_asan",
+            ]
+            + _pybind_includes()
+            + [
+                f"-I{ROCM}/include",
+                src,
+                f"-L{ROCM}/lib",
+                "-lamd_smi",
+                f"-Wl,-rpath,{ROCM}/lib",
+                "-o",
+                out,
+            ]
+        )
+    return out
+
+
 def build_hiphealth(force: bool = False) -> str:
     src = os.path.join(CPP_DIR, "hiphealth.hip")
     out = os.path.join(PKG_DIR, f"_hiphealth{_ext_suffix()}")

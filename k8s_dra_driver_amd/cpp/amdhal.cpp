@@ -263,7 +263,13 @@ py::dict hal_lib_version() {
 
 }  // namespace
 
-PYBIND11_MODULE(_amdhal, m) {
+// PYBIND11_MODULE_NAME lets the ASan CI flavor build this same source as
+// `_amdhal_asan` (module init name must match the .so filename).
+#ifndef PYBIND11_MODULE_NAME
+#define PYBIND11_MODULE_NAME _amdhal
+#endif
+
+PYBIND11_MODULE(PYBIND11_MODULE_NAME, m) {
   m.doc() = "AMD SMI native binding for the MI355X DRA driver";
   m.def("init", &hal_init, "Initialize AMD SMI (idempotent)");
   m.def("shutdown", &hal_shutdown, "Shut down AMD SMI (idempotent)");

@@ -34,7 +34,7 @@ import ssl
 import subprocess
 import tempfile
 import threading
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Dict, List, Optional, Tuple
 
 import yaml
 

@@ -3,7 +3,6 @@
 import pytest
 
 from k8s_dra_driver_amd.allocator.structured import AllocationError, Allocator
-from k8s_dra_driver_amd.hal import FakeDeviceLib, FakeNodeConfig
 from k8s_dra_driver_amd.hal.model import AllocatableDevice
 from k8s_dra_driver_amd.topology.xgmi import (
     is_fully_connected,

@@ -6,7 +6,6 @@ match cpp/amdhal.cpp::describe_processor)."""
 
 from typing import List
 
-import pytest
 
 from k8s_dra_driver_amd.hal.amdsmi import AmdSmiDeviceLib
 

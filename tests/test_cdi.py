@@ -15,7 +15,6 @@ from k8s_dra_driver_amd.cdi.spec import (
     read_spec_file,
     write_spec_file,
 )
-from k8s_dra_driver_amd.hal import FakeDeviceLib
 from k8s_dra_driver_amd.hal.model import AllocatableDevice
 
 

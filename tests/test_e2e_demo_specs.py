@@ -11,7 +11,7 @@ import pytest
 import yaml
 
 from k8s_dra_driver_amd import DRIVER_NAME
-from k8s_dra_driver_amd.allocator.structured import AllocationError, Allocator
+from k8s_dra_driver_amd.allocator.structured import Allocator
 from k8s_dra_driver_amd.hal import FakeDeviceLib
 from k8s_dra_driver_amd.kube.client import InMemoryKube
 from k8s_dra_driver_amd.plugin.driver import ClaimRef, Driver

@@ -6,7 +6,7 @@ hardware-free backend) so its semantics are pinned here.
 
 import pytest
 
-from k8s_dra_driver_amd.hal import FakeDeviceLib, FakeNodeConfig
+from k8s_dra_driver_amd.hal import FakeDeviceLib
 from k8s_dra_driver_amd.hal.base import HalError
 from k8s_dra_driver_amd.hal.fake import FaultInjector
 from k8s_dra_driver_amd.hal.model import (

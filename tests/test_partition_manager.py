@@ -2,7 +2,6 @@
 
 import pytest
 
-from k8s_dra_driver_amd.hal import FakeDeviceLib
 from k8s_dra_driver_amd.partition.manager import PartitionManager, RepartitionRefused
 
 

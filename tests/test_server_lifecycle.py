@@ -4,7 +4,6 @@ import os
 
 import grpc
 
-from k8s_dra_driver_amd import DRIVER_NAME
 from k8s_dra_driver_amd.hal import FakeDeviceLib
 from k8s_dra_driver_amd.kube.client import InMemoryKube
 from k8s_dra_driver_amd.plugin.driver import Driver

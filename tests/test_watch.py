@@ -1,11 +1,8 @@
 """Watch/informer tests: in-memory hooks, HTTP streaming watch, and the
 controller's prompt allocation on claim arrival."""
 
-import json
-import threading
 import time
 
-import pytest
 
 from k8s_dra_driver_amd import DRIVER_NAME
 from k8s_dra_driver_amd.controller.manager import ControllerManager

@@ -42,6 +42,13 @@ class KubeClient(abc.ABC):
         """DeviceClass objects (controller allocator input); optional."""
         return []
 
+    def resource_api_versions(self) -> List[str]:
+        """Served versions of the resource.k8s.io group, preferred first
+        (apiserver discovery). Drives version-negotiated ResourceSlice
+        publication (reference draplugin.go:342-350 multi-version
+        pattern; K8s 1.33 flattens the Device shape in v1beta2)."""
+        return ["v1beta1"]
+
     def create_event(self, namespace: str, event: dict) -> None:
         """Emit a Kubernetes Event (operator visibility); best-effort."""
 
@@ -85,6 +92,9 @@ class InMemoryKube(KubeClient):
 
     def __init__(self):
         self._lock = threading.RLock()
+        #: served resource.k8s.io versions, preferred first (tests flip
+        #: this to ["v1beta2", "v1beta1"] to exercise negotiation)
+        self.api_versions: List[str] = ["v1beta1"]
         self.resource_claims: Dict[str, dict] = {}  # key: ns/name
         self.resource_slices: Dict[str, dict] = {}
         self.nodes: Dict[str, dict] = {}
@@ -150,6 +160,9 @@ class InMemoryKube(KubeClient):
         with self._lock:
             self.device_classes[obj["metadata"]["name"]] = copy.deepcopy(obj)
             return obj
+
+    def resource_api_versions(self) -> List[str]:
+        return list(self.api_versions)
 
     def create_event(self, namespace: str, event: dict) -> None:
         with self._lock:

@@ -29,7 +29,6 @@ from .client import Conflict, KubeClient, NotFound
 
 log = logging.getLogger(__name__)
 
-RESOURCE_V1BETA1 = "/apis/resource.k8s.io/v1beta1"
 CORE_V1 = "/api/v1"
 
 
@@ -80,6 +79,7 @@ class HttpKube(KubeClient):
         self._client_lock = threading.Lock()
         self._client_epoch = -1
         self._client: Optional[httpx.Client] = None
+        self._resource_base: Optional[str] = None
         self._http()  # fail fast on bad TLS material
 
     def _http(self) -> httpx.Client:
@@ -130,22 +130,53 @@ class HttpKube(KubeClient):
         r.raise_for_status()
         return r.json() if r.content else {}
 
+    # -- version negotiation ------------------------------------------------
+    def resource_api_versions(self) -> List[str]:
+        """Served resource.k8s.io versions, preferred first (apiserver
+        group discovery; falls back to v1beta1 when discovery fails)."""
+        try:
+            out = self._req("GET", "/apis/resource.k8s.io")
+            versions = [
+                v.get("version")
+                for v in out.get("versions", [])
+                if v.get("version")
+            ]
+            pref = (out.get("preferredVersion") or {}).get("version")
+            if pref and pref in versions:
+                versions.remove(pref)
+                versions.insert(0, pref)
+            return versions or ["v1beta1"]
+        except Exception:
+            return ["v1beta1"]
+
+    @property
+    def _rbase(self) -> str:
+        """REST base for the resource group, negotiated once (newest
+        mutually-supported of v1beta2/v1beta1)."""
+        if self._resource_base is None:
+            served = self.resource_api_versions()
+            pick = next(
+                (v for v in ("v1beta2", "v1beta1") if v in served), "v1beta1"
+            )
+            self._resource_base = f"/apis/resource.k8s.io/{pick}"
+        return self._resource_base
+
     # -- KubeClient --------------------------------------------------------
     def get_resource_claim(self, namespace: str, name: str) -> dict:
         return self._req(
-            "GET", f"{RESOURCE_V1BETA1}/namespaces/{namespace}/resourceclaims/{name}"
+            "GET", f"{self._rbase}/namespaces/{namespace}/resourceclaims/{name}"
         )
 
     def create_resource_slice(self, obj: dict) -> dict:
-        return self._req("POST", f"{RESOURCE_V1BETA1}/resourceslices", obj)
+        return self._req("POST", f"{self._rbase}/resourceslices", obj)
 
     def update_resource_slice(self, obj: dict) -> dict:
         name = obj["metadata"]["name"]
-        return self._req("PUT", f"{RESOURCE_V1BETA1}/resourceslices/{name}", obj)
+        return self._req("PUT", f"{self._rbase}/resourceslices/{name}", obj)
 
     def delete_resource_slice(self, name: str) -> None:
         try:
-            self._req("DELETE", f"{RESOURCE_V1BETA1}/resourceslices/{name}")
+            self._req("DELETE", f"{self._rbase}/resourceslices/{name}")
         except NotFound:
             pass
 
@@ -153,11 +184,11 @@ class HttpKube(KubeClient):
         params = ""
         if driver:
             params = f"?fieldSelector=spec.driver%3D{driver}"
-        out = self._req("GET", f"{RESOURCE_V1BETA1}/resourceslices{params}")
+        out = self._req("GET", f"{self._rbase}/resourceslices{params}")
         return out.get("items", [])
 
     def list_resource_claims(self) -> List[dict]:
-        out = self._req("GET", f"{RESOURCE_V1BETA1}/resourceclaims")
+        out = self._req("GET", f"{self._rbase}/resourceclaims")
         return out.get("items", [])
 
     def update_resource_claim_status(self, obj: dict) -> dict:
@@ -165,12 +196,12 @@ class HttpKube(KubeClient):
         name = obj["metadata"]["name"]
         return self._req(
             "PUT",
-            f"{RESOURCE_V1BETA1}/namespaces/{ns}/resourceclaims/{name}/status",
+            f"{self._rbase}/namespaces/{ns}/resourceclaims/{name}/status",
             obj,
         )
 
     def get_device_classes(self) -> List[dict]:
-        out = self._req("GET", f"{RESOURCE_V1BETA1}/deviceclasses")
+        out = self._req("GET", f"{self._rbase}/deviceclasses")
         return out.get("items", [])
 
     # -- watches -----------------------------------------------------------
@@ -262,12 +293,12 @@ class HttpKube(KubeClient):
 
     def watch_resource_claims(self, handler):
         return self._watch(
-            f"{RESOURCE_V1BETA1}/resourceclaims", handler, name="claims-watch"
+            f"{self._rbase}/resourceclaims", handler, name="claims-watch"
         )
 
     def watch_resource_slices(self, handler):
         return self._watch(
-            f"{RESOURCE_V1BETA1}/resourceslices", handler, name="slices-watch"
+            f"{self._rbase}/resourceslices", handler, name="slices-watch"
         )
 
     def create_event(self, namespace: str, event: dict) -> None:

@@ -19,11 +19,12 @@ from typing import Optional
 from .client import Conflict, InMemoryKube, NotFound
 
 _CLAIM_RE = re.compile(
-    r"^/apis/resource\.k8s\.io/v1beta1/namespaces/([^/]+)/resourceclaims/([^/]+?)(/status)?$"
+    r"^/apis/resource\.k8s\.io/v1beta[12]/namespaces/([^/]+)/resourceclaims/([^/]+?)(/status)?$"
 )
-_SLICE_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta1/resourceslices(?:/([^/]+))?$")
-_CLAIMS_ALL = "/apis/resource.k8s.io/v1beta1/resourceclaims"
-_CLASSES = "/apis/resource.k8s.io/v1beta1/deviceclasses"
+_SLICE_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta[12]/resourceslices(?:/([^/]+))?$")
+_CLAIMS_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta[12]/resourceclaims$")
+_CLASSES_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta[12]/deviceclasses$")
+_GROUP_DISCOVERY = "/apis/resource.k8s.io"
 _NODE_RE = re.compile(r"^/api/v1/nodes/([^/]+)$")
 
 
@@ -59,7 +60,27 @@ class MiniApiServer:
 
             def do_GET(self):
                 path = self.path.split("?")[0]
-                if path == _CLAIMS_ALL and "watch=true" in self.path:
+                if path == _GROUP_DISCOVERY:
+                    versions = outer.store.api_versions
+                    return self._json(
+                        200,
+                        {
+                            "kind": "APIGroup",
+                            "name": "resource.k8s.io",
+                            "versions": [
+                                {
+                                    "groupVersion": f"resource.k8s.io/{v}",
+                                    "version": v,
+                                }
+                                for v in versions
+                            ],
+                            "preferredVersion": {
+                                "groupVersion": f"resource.k8s.io/{versions[0]}",
+                                "version": versions[0],
+                            },
+                        },
+                    )
+                if _CLAIMS_RE.match(path) and "watch=true" in self.path:
                     return self._watch_kind("ResourceClaim")
                 m = _CLAIM_RE.match(path)
                 if m:
@@ -69,9 +90,9 @@ class MiniApiServer:
                         )
                     except NotFound:
                         return self._json(404, {"reason": "NotFound"})
-                if path == _CLAIMS_ALL:
+                if _CLAIMS_RE.match(path):
                     return self._list(outer.store.list_resource_claims())
-                if path == _CLASSES:
+                if _CLASSES_RE.match(path):
                     return self._list(outer.store.get_device_classes())
                 m = _SLICE_RE.match(path)
                 if m and not m.group(1):

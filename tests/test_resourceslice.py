@@ -184,3 +184,50 @@ class TestDriftSelfHeal:
         kube.delete_resource_slice(name)
         pub.publish(devs)  # same fingerprint, but observed state drifted
         assert len(kube.list_resource_slices("gpu.amd.com")) == 1
+
+
+class TestV1beta2Publication:
+    """Version-negotiated publication: one code path,
+    flattened Device shape when the apiserver serves v1beta2."""
+
+    def _basic_dev(self):
+        return {
+            "name": "gpu-0",
+            "basic": {
+                "attributes": {"gpu.amd.com/type": {"string": "gpu"}},
+                "capacity": {"gpu.amd.com/memory": {"value": "288Gi"}},
+            },
+        }
+
+    def test_v1beta1_default_keeps_basic_shape(self):
+        kube = InMemoryKube()
+        pub = _pub(kube)
+        pub.publish([self._basic_dev()])
+        s = kube.list_resource_slices("gpu.amd.com")[0]
+        assert s["apiVersion"] == "resource.k8s.io/v1beta1"
+        assert "basic" in s["spec"]["devices"][0]
+
+    def test_v1beta2_flattens_devices(self):
+        kube = InMemoryKube()
+        kube.api_versions = ["v1beta2", "v1beta1"]
+        pub = _pub(kube)
+        pub.publish([self._basic_dev()])
+        s = kube.list_resource_slices("gpu.amd.com")[0]
+        assert s["apiVersion"] == "resource.k8s.io/v1beta2"
+        d = s["spec"]["devices"][0]
+        assert "basic" not in d
+        assert d["attributes"]["gpu.amd.com/type"] == {"string": "gpu"}
+        assert d["capacity"]["gpu.amd.com/memory"] == {"value": "288Gi"}
+
+    def test_v1beta2_self_heal_keeps_flat_shape(self):
+        kube = InMemoryKube()
+        kube.api_versions = ["v1beta2"]
+        pub = _pub(kube)
+        pub.publish([self._basic_dev()])
+        assert pub.start_self_heal()
+        name = kube.list_resource_slices("gpu.amd.com")[0]["metadata"]["name"]
+        kube.delete_resource_slice(name)
+        s = kube.list_resource_slices("gpu.amd.com")[0]
+        assert s["apiVersion"] == "resource.k8s.io/v1beta2"
+        assert "basic" not in s["spec"]["devices"][0]
+        pub.stop_self_heal()

@@ -91,3 +91,33 @@ def test_controller_allocates_promptly_on_watch():
         )
     finally:
         mgr.stop()
+
+
+def test_http_kube_negotiates_v1beta2(tmp_path):
+    """End-to-end version negotiation over the wire: apiserver serves
+    v1beta2 preferred -> HttpKube uses the v1beta2 REST base and the
+    publisher emits flattened devices through it."""
+    srv = MiniApiServer().start()
+    srv.store.api_versions = ["v1beta2", "v1beta1"]
+    kc = srv.write_kubeconfig(str(tmp_path / "kc"))
+    client = HttpKube(kubeconfig=kc, qps=1000, burst=1000)
+    assert client.resource_api_versions()[0] == "v1beta2"
+    pub = ResourceSlicePublisher(
+        client, driver_name=DRIVER_NAME, node_name="vn"
+    )
+    pub.publish(
+        [{"name": "gpu-0", "basic": {"attributes": {}, "capacity": {}}}]
+    )
+    s = srv.store.list_resource_slices(DRIVER_NAME)[0]
+    assert s["apiVersion"] == "resource.k8s.io/v1beta2"
+    assert "basic" not in s["spec"]["devices"][0]
+    srv.stop()
+
+
+def test_http_kube_defaults_to_v1beta1(tmp_path):
+    srv = MiniApiServer().start()
+    kc = srv.write_kubeconfig(str(tmp_path / "kc"))
+    client = HttpKube(kubeconfig=kc, qps=1000, burst=1000)
+    assert client.resource_api_versions() == ["v1beta1"]
+    assert client._rbase.endswith("/v1beta1")
+    srv.stop()

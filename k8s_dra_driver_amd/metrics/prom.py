@@ -77,6 +77,11 @@ class PluginMetrics:
             "Dynamic partition mode switches performed",
             registry=self.registry,
         )
+        self.isolation_violations = Counter(
+            "dra_shared_isolation_violations_total",
+            "Shared-GPU processes caught with stripped/altered CU masks",
+            registry=self.registry,
+        )
 
     @contextlib.contextmanager
     def time_prepare(self):

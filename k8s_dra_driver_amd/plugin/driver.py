@@ -62,6 +62,7 @@ class Driver:
         max_concurrent_claims: int = 16,
         metrics: Optional[PluginMetrics] = None,
         device_kinds: Optional[List[str]] = None,
+        shared_enforcement: str = "warn",  # off | warn | kill
     ):
         self.lib = lib
         self.kube = kube
@@ -111,6 +112,34 @@ class Driver:
         self.health = HealthMonitor(
             lib, on_change=lambda _unhealthy: self.publish_resources()
         )
+        # Out-of-band shared-GPU isolation enforcement (sharing.go:211-221
+        # analog): detect containers that stripped/altered their CU mask.
+        self.enforcer = None
+        if shared_enforcement != "off":
+            from ..sharing.enforce import SharedEnforcer
+
+            self.enforcer = SharedEnforcer(
+                shared,
+                action=shared_enforcement,
+                on_violation=self._on_isolation_violation,
+            )
+
+    def _on_isolation_violation(self, v) -> None:
+        """Surface a shared-GPU isolation violation: metric + Warning
+        event on the offending claim (namespace/name from its checkpoint)."""
+        self.metrics.isolation_violations.inc()
+        pc = self.state.checkpoints.read(v.claim_uid) if v.claim_uid else None
+        info = ClaimRef(
+            namespace=pc.namespace if pc else "default",
+            name=pc.name if pc else v.claim_uid,
+            uid=v.claim_uid,
+        )
+        self._emit_claim_warning(
+            info,
+            "SharedIsolationViolation",
+            f"pid {v.pid}: {v.kind} — {v.detail}"
+            + (" (process killed)" if self.enforcer.action == "kill" else ""),
+        )
 
     # ------------------------------------------------------------------
     def startup(self) -> None:
@@ -123,6 +152,8 @@ class Driver:
         self.publisher.start_self_heal()
 
     def shutdown(self, unpublish: bool = True) -> None:
+        if self.enforcer is not None:
+            self.enforcer.stop()
         self.publisher.stop_self_heal()
         self._pool.shutdown(wait=True)
         self.state.close()

@@ -86,6 +86,14 @@ def build_parser() -> argparse.ArgumentParser:
         default=int(_env("METRICS_PORT", "0")),
         help="serve Prometheus /metrics on this port (0 = disabled)",
     )
+    p.add_argument(
+        "--shared-enforcement",
+        default=_env("SHARED_ENFORCEMENT", "warn"),
+        choices=["off", "warn", "kill"],
+        help="shared-GPU isolation enforcement for containers that strip "
+        "or alter their CU mask: warn = Warning event + metric, kill = "
+        "SIGKILL the offending process (sharing.go:211-221 parity)",
+    )
     p.add_argument("-v", "--verbosity", type=int, default=int(_env("LOG_LEVEL", "1")))
     p.add_argument(
         "--logging-format",
@@ -133,9 +141,12 @@ def main(argv=None) -> int:
         checkpoint_root=os.path.join(args.plugin_path, "state"),
         metrics=metrics,
         device_kinds=[s.strip() for s in args.device_classes.split(",") if s.strip()],
+        shared_enforcement=args.shared_enforcement,
     )
     driver.startup()
     driver.health.start()  # failure detection -> slice self-healing
+    if driver.enforcer is not None:
+        driver.enforcer.start()  # shared-GPU isolation watchdog
     server = PluginServer(
         driver,
         plugin_dir=args.plugin_path,

@@ -265,3 +265,115 @@ class TestRcclFabric:
         assert isinstance(cap, bool)
         g0 = real_lib.enumerate()[0]
         assert g0.repartition_capable == cap
+
+
+class TestSharedEnforcementOnHardware:
+    """Adversarial case on real KFD: a process attached to
+    the GPU whose env claims a SharedCompute session but carries no CU
+    mask must be detected via /sys/class/kfd/kfd/proc attribution."""
+
+    def test_stripped_mask_process_detected(self, tmp_path):
+        import subprocess
+        import sys
+        import time as _time
+
+        from k8s_dra_driver_amd.api.types import SharedComputeSettings
+        from k8s_dra_driver_amd.hal.amdsmi import AmdSmiDeviceLib
+        from k8s_dra_driver_amd.hal.model import AllocatableDevice
+        from k8s_dra_driver_amd.sharing.enforce import SharedEnforcer
+        from k8s_dra_driver_amd.sharing.shared import SharedComputeManager
+
+        lib = AmdSmiDeviceLib()
+        lib.open()
+        try:
+            mgr = SharedComputeManager(
+                root=str(tmp_path / "shared"), use_tmpfs=False
+            )
+            session = mgr.start_session(
+                "hw-claim-1",
+                [AllocatableDevice.from_gpu(lib.enumerate()[0])],
+                SharedComputeSettings(default_cu_share_percent=25),
+            )
+            # adversarial pod process: opens the GPU, claims the session,
+            # but scrubbed HSA_CU_MASK from its environment
+            env = dict(os.environ)
+            env.pop("HSA_CU_MASK", None)
+            env["AMD_DRA_SHARED_SESSION"] = session.session_id
+            env["AMD_DRA_CLAIM_UID"] = "hw-claim-1"
+            code = (
+                "import torch, time;"
+                "torch.zeros(4, device='cuda:0');"
+                "print('attached', flush=True);"
+                "time.sleep(120)"
+            )
+            proc = subprocess.Popen(
+                [sys.executable, "-c", code],
+                env=env,
+                stdout=subprocess.PIPE,
+                text=True,
+            )
+            try:
+                assert proc.stdout.readline().strip() == "attached"
+                _time.sleep(1.0)  # KFD proc entry settles
+                enf = SharedEnforcer(mgr)
+                assert proc.pid in enf.gpu_pids(), (
+                    f"pid {proc.pid} not in KFD proc list {enf.gpu_pids()}"
+                )
+                violations = enf.scan()
+                mine = [v for v in violations if v.pid == proc.pid]
+                assert len(mine) == 1 and mine[0].kind == "stripped", violations
+            finally:
+                proc.kill()
+                proc.wait()
+        finally:
+            lib.close()
+
+    def test_compliant_process_not_flagged(self, tmp_path):
+        import subprocess
+        import sys
+        import time as _time
+
+        from k8s_dra_driver_amd.api.types import SharedComputeSettings
+        from k8s_dra_driver_amd.hal.amdsmi import AmdSmiDeviceLib
+        from k8s_dra_driver_amd.hal.model import AllocatableDevice
+        from k8s_dra_driver_amd.sharing.enforce import SharedEnforcer
+        from k8s_dra_driver_amd.sharing.shared import SharedComputeManager
+
+        lib = AmdSmiDeviceLib()
+        lib.open()
+        try:
+            mgr = SharedComputeManager(
+                root=str(tmp_path / "shared"), use_tmpfs=False
+            )
+            session = mgr.start_session(
+                "hw-claim-2",
+                [AllocatableDevice.from_gpu(lib.enumerate()[0])],
+                SharedComputeSettings(default_cu_share_percent=25),
+            )
+            env = dict(os.environ)
+            for e in session.env:
+                k, _, v = e.partition("=")
+                env[k] = v
+            env["AMD_DRA_CLAIM_UID"] = "hw-claim-2"
+            code = (
+                "import torch, time;"
+                "torch.zeros(4, device='cuda:0');"
+                "print('attached', flush=True);"
+                "time.sleep(120)"
+            )
+            proc = subprocess.Popen(
+                [sys.executable, "-c", code],
+                env=env,
+                stdout=subprocess.PIPE,
+                text=True,
+            )
+            try:
+                assert proc.stdout.readline().strip() == "attached"
+                _time.sleep(1.0)
+                violations = SharedEnforcer(mgr).scan()
+                assert [v for v in violations if v.pid == proc.pid] == []
+            finally:
+                proc.kill()
+                proc.wait()
+        finally:
+            lib.close()

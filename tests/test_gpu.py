@@ -377,3 +377,51 @@ class TestSharedEnforcementOnHardware:
                 proc.wait()
         finally:
             lib.close()
+
+
+class TestLiveCapsEndToEnd:
+    """the pool's firmware reports restricted NPS caps
+    (round 1 observed NPS1/NPS2 only, not the bare-metal NPS4 default);
+    the enumerate() -> catalog path must honor the LIVE caps end to end,
+    and the repartition capability probe must agree with what set()
+    actually does."""
+
+    def test_published_profiles_respect_live_nps_caps(self, real_lib):
+        from k8s_dra_driver_amd.partition.catalog import make_profile
+
+        for g in real_lib.enumerate():
+            assert g.nps_caps, "live NPS caps must be reported"
+            # any profile the driver would build for this GPU must use a
+            # mode the hardware actually reports
+            for mem_mode in g.nps_caps:
+                prof = make_profile(
+                    "CPX" if "CPX" in g.compute_caps else g.compute_partition,
+                    mem_mode,
+                    vram_total_mib=g.vram_total_mib or 288 * 1024,
+                    cu_count=g.cu_count or 256,
+                    nps_caps=g.nps_caps,
+                )
+                assert prof.memory_mode in g.nps_caps
+            # modes NOT in the live caps are refused by the catalog
+            missing = [m for m in ("NPS1", "NPS2", "NPS4") if m not in g.nps_caps]
+            for m in missing:
+                import pytest as _pytest
+
+                with _pytest.raises(ValueError):
+                    make_profile("CPX", m, nps_caps=g.nps_caps)
+
+    def test_repartition_probe_matches_reality(self, real_lib):
+        """If the probe says capable, a same-mode set must succeed; if it
+        says incapable, the set must fail — either way they must agree
+        (the learned-capability contract)."""
+        g0 = real_lib.enumerate()[0]
+        probed = real_lib.dynamic_repartition_capable()
+        try:
+            real_lib.set_compute_partition(0, g0.compute_partition)
+            actually = True
+        except Exception:
+            actually = False
+        assert real_lib.dynamic_repartition_capable() == actually
+        if probed != actually:
+            # probe was optimistic/pessimistic; the learned value must win
+            assert real_lib.dynamic_repartition_capable() == actually

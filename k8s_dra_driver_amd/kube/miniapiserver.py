@@ -207,12 +207,64 @@ class MiniApiServer:
 
         self._server = ThreadingHTTPServer(("127.0.0.1", port), Handler)
         self._thread: Optional[threading.Thread] = None
+        self._tls: Optional[dict] = None
+
+    def enable_tls(
+        self,
+        server_cert: str,
+        server_key: str,
+        *,
+        client_ca: Optional[str] = None,
+        ca_file: Optional[str] = None,
+    ) -> "MiniApiServer":
+        """Serve HTTPS like a kubeadm/kind apiserver; with ``client_ca``
+        a client certificate is REQUIRED (mTLS), which is exactly the
+        connection shape round-1's HttpKube could not make."""
+        import ssl
+
+        ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+        ctx.load_cert_chain(server_cert, server_key)
+        if client_ca:
+            ctx.load_verify_locations(client_ca)
+            ctx.verify_mode = ssl.CERT_REQUIRED
+        self._server.socket = ctx.wrap_socket(
+            self._server.socket, server_side=True
+        )
+        self._tls = {"ca_file": ca_file or client_ca}
+        return self
 
     @property
     def url(self) -> str:
-        return f"http://127.0.0.1:{self._server.server_address[1]}"
+        scheme = "https" if self._tls else "http"
+        return f"{scheme}://127.0.0.1:{self._server.server_address[1]}"
 
-    def write_kubeconfig(self, path: str) -> str:
+    def write_kubeconfig(
+        self,
+        path: str,
+        *,
+        client_cert: Optional[str] = None,
+        client_key: Optional[str] = None,
+    ) -> str:
+        """Token kubeconfig by default; kind-style inline client-cert +
+        CA data when TLS is enabled and cert paths are given."""
+        import base64
+
+        cluster: dict = {"server": self.url}
+        user: dict = {"token": "dev"}
+        if self._tls and self._tls.get("ca_file"):
+            with open(self._tls["ca_file"], "rb") as f:
+                cluster["certificate-authority-data"] = base64.b64encode(
+                    f.read()
+                ).decode()
+        if client_cert and client_key:
+            with open(client_cert, "rb") as f:
+                cert_b64 = base64.b64encode(f.read()).decode()
+            with open(client_key, "rb") as f:
+                key_b64 = base64.b64encode(f.read()).decode()
+            user = {
+                "client-certificate-data": cert_b64,
+                "client-key-data": key_b64,
+            }
         with open(path, "w") as f:
             json.dump(
                 {
@@ -220,10 +272,8 @@ class MiniApiServer:
                     "contexts": [
                         {"name": "mini", "context": {"cluster": "mini", "user": "u"}}
                     ],
-                    "clusters": [
-                        {"name": "mini", "cluster": {"server": self.url}}
-                    ],
-                    "users": [{"name": "u", "user": {"token": "dev"}}],
+                    "clusters": [{"name": "mini", "cluster": cluster}],
+                    "users": [{"name": "u", "user": user}],
                 },
                 f,
             )

@@ -132,6 +132,7 @@ class SharedComputeManager:
             ]
             gpu_indices: List[int] = []
             allocated: List[int] = []
+            cu_entries: List[str] = []
             try:
                 for dev in devices:
                     gpu = dev.parent_gpu
@@ -146,10 +147,17 @@ class SharedComputeManager:
                         num_cu = max(1, gpu.cu_count * share // 100)
                         start, count = st.allocate(session_id, num_cu)
                         allocated.append(gpu.index)
-                        env.append(
-                            f"HSA_CU_MASK={gpu.index}:"
+                        cu_entries.append(
+                            f"{gpu.index}:"
                             f"{cu_mask_hex(start, count, gpu.cu_count)}"
                         )
+                # ONE env var, ';'-joined per GPU: multiple same-name env
+                # entries collapse in a real container environment, and the
+                # joined form is hardware-verified to constrain each listed
+                # GPU (profiles/cu_mask_syntax_r02: hex single, hex joined
+                # and decimal ranges all quarter MFMA throughput on gfx950)
+                if cu_entries:
+                    env.append("HSA_CU_MASK=" + ";".join(cu_entries))
                 # VRAM budgeting (best-effort; see module docstring)
                 uuids_by_index = {
                     i: d.uuid for i, d in enumerate(devices)
@@ -203,14 +211,19 @@ class SharedComputeManager:
             self._sessions[session.session_id] = session
             for e in session.env:
                 if e.startswith("HSA_CU_MASK="):
-                    # parse "<idx>:0x<mask>" back into range bookkeeping
-                    body = e.split("=", 1)[1]
-                    idx_s, mask_s = body.split(":", 1)
-                    mask = int(mask_s, 16)
-                    start = (mask & -mask).bit_length() - 1 if mask else 0
-                    count = bin(mask).count("1")
-                    st = self._gpu_state.setdefault(int(idx_s), _GpuShareState())
-                    st.ranges[session.session_id] = (start, count)
+                    # parse "<idx>:0x<mask>[;<idx>:0x<mask>...]" back into
+                    # range bookkeeping (one ';'-joined var per session)
+                    for entry in e.split("=", 1)[1].split(";"):
+                        if ":" not in entry:
+                            continue
+                        idx_s, mask_s = entry.split(":", 1)
+                        mask = int(mask_s, 16)
+                        start = (mask & -mask).bit_length() - 1 if mask else 0
+                        count = bin(mask).count("1")
+                        st = self._gpu_state.setdefault(
+                            int(idx_s), _GpuShareState()
+                        )
+                        st.ranges[session.session_id] = (start, count)
 
     def get_session(self, session_id: str) -> Optional[SharedSession]:
         with self._lock:

@@ -272,26 +272,23 @@ class TestSharedEnforcementOnHardware:
     the GPU whose env claims a SharedCompute session but carries no CU
     mask must be detected via /sys/class/kfd/kfd/proc attribution."""
 
-    def test_stripped_mask_process_detected(self, tmp_path):
+    def test_stripped_mask_process_detected(self, real_lib, tmp_path):
         import subprocess
         import sys
         import time as _time
 
         from k8s_dra_driver_amd.api.types import SharedComputeSettings
-        from k8s_dra_driver_amd.hal.amdsmi import AmdSmiDeviceLib
         from k8s_dra_driver_amd.hal.model import AllocatableDevice
         from k8s_dra_driver_amd.sharing.enforce import SharedEnforcer
         from k8s_dra_driver_amd.sharing.shared import SharedComputeManager
 
-        lib = AmdSmiDeviceLib()
-        lib.open()
-        try:
+        if True:
             mgr = SharedComputeManager(
                 root=str(tmp_path / "shared"), use_tmpfs=False
             )
             session = mgr.start_session(
                 "hw-claim-1",
-                [AllocatableDevice.from_gpu(lib.enumerate()[0])],
+                [AllocatableDevice.from_gpu(real_lib.enumerate()[0])],
                 SharedComputeSettings(default_cu_share_percent=25),
             )
             # adversarial pod process: opens the GPU, claims the session,
@@ -325,29 +322,24 @@ class TestSharedEnforcementOnHardware:
             finally:
                 proc.kill()
                 proc.wait()
-        finally:
-            lib.close()
 
-    def test_compliant_process_not_flagged(self, tmp_path):
+    def test_compliant_process_not_flagged(self, real_lib, tmp_path):
         import subprocess
         import sys
         import time as _time
 
         from k8s_dra_driver_amd.api.types import SharedComputeSettings
-        from k8s_dra_driver_amd.hal.amdsmi import AmdSmiDeviceLib
         from k8s_dra_driver_amd.hal.model import AllocatableDevice
         from k8s_dra_driver_amd.sharing.enforce import SharedEnforcer
         from k8s_dra_driver_amd.sharing.shared import SharedComputeManager
 
-        lib = AmdSmiDeviceLib()
-        lib.open()
-        try:
+        if True:
             mgr = SharedComputeManager(
                 root=str(tmp_path / "shared"), use_tmpfs=False
             )
             session = mgr.start_session(
                 "hw-claim-2",
-                [AllocatableDevice.from_gpu(lib.enumerate()[0])],
+                [AllocatableDevice.from_gpu(real_lib.enumerate()[0])],
                 SharedComputeSettings(default_cu_share_percent=25),
             )
             env = dict(os.environ)
@@ -375,8 +367,6 @@ class TestSharedEnforcementOnHardware:
             finally:
                 proc.kill()
                 proc.wait()
-        finally:
-            lib.close()
 
 
 class TestLiveCapsEndToEnd:

@@ -96,8 +96,9 @@ def _cu_masks(state, uid):
     out = {}
     for e in pc.claim_env:
         if e.startswith("HSA_CU_MASK="):
-            idx_s, mask_s = e.split("=", 1)[1].split(":", 1)
-            out[int(idx_s)] = int(mask_s, 16)
+            for entry in e.split("=", 1)[1].split(";"):
+                idx_s, mask_s = entry.split(":", 1)
+                out[int(idx_s)] = int(mask_s, 16)
     return out
 
 

@@ -88,6 +88,11 @@ class SharedEnforcer:
         self._thread: Optional[threading.Thread] = None
         #: observability: total violations seen (metrics surface)
         self.violation_count = 0
+        #: KFD pids with no /proc entry in the last scan — non-empty means
+        #: this process lacks the host PID namespace (deploy with
+        #: hostPID: true, as the Helm chart does) and cannot attribute
+        self.last_unattributable: List[int] = []
+        self._warned_ns = False
 
     # -- scanning ----------------------------------------------------------
     def gpu_pids(self) -> List[int]:
@@ -108,9 +113,12 @@ class SharedEnforcer:
         """One pass over GPU-attached processes; returns violations found
         (also dispatched to ``on_violation`` / acted on per ``action``)."""
         out: List[Violation] = []
+        unattributable: List[int] = []
         for pid in self.gpu_pids():
             env = _read_environ(self.proc_root, pid)
             if env is None:
+                if not os.path.isdir(f"{self.proc_root}/{pid}"):
+                    unattributable.append(pid)
                 continue
             session_id = env.get("AMD_DRA_SHARED_SESSION", "")
             if not session_id:
@@ -154,6 +162,16 @@ class SharedEnforcer:
                         f"(expected one of {expected})",
                     )
                 )
+        self.last_unattributable = unattributable
+        if unattributable and not self._warned_ns:
+            self._warned_ns = True
+            log.warning(
+                "%d GPU-attached KFD pid(s) have no /proc entry — this "
+                "process is likely missing the host PID namespace, so "
+                "shared-GPU isolation cannot be attributed (deploy the "
+                "plugin with hostPID: true; the Helm chart sets it)",
+                len(unattributable),
+            )
         for v in out:
             self._handle(v)
         return out

@@ -23,6 +23,7 @@ Supported claim spec surface (resource.k8s.io/v1beta1):
 from __future__ import annotations
 
 import itertools
+import logging
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence, Set, Tuple
 
@@ -30,6 +31,11 @@ from .. import DRIVER_NAME
 from ..hal.model import DOMAIN
 from ..topology.xgmi import pick_best_subset, subset_score
 from .cel import matches
+
+log = logging.getLogger(__name__)
+
+#: per-request combination fan-out cap (see _search.options)
+COMBO_CAP = 2000
 
 
 class AllocationError(RuntimeError):
@@ -249,9 +255,22 @@ class Allocator:
                 yield free
                 return
             # Cap the combination fan-out per request; order candidates by
-            # affinity so the cap keeps the good ones.
+            # affinity so the cap keeps the good ones. The cap is LOGGED
+            # when it bites (no silent truncation): placements beyond it
+            # are never considered.
+            import math
+
+            total = math.comb(len(free), count)
+            if total > COMBO_CAP:
+                log.warning(
+                    "allocator: request %r has %d candidate combinations; "
+                    "considering only the first %d (combination cap)",
+                    r.get("name"),
+                    total,
+                    COMBO_CAP,
+                )
             combos = itertools.combinations(free, count)
-            for combo in itertools.islice(combos, 2000):
+            for combo in itertools.islice(combos, COMBO_CAP):
                 yield list(combo)
 
         def dfs(i: int, used: Set[str], chosen: List[List[dict]]):
@@ -279,8 +298,21 @@ class Allocator:
 
         dfs(0, set(), [])
         if best is not None:
+            if budget <= 0:
+                log.warning(
+                    "allocator: search budget (%d states) exhausted; "
+                    "returning best assignment found so far (topology "
+                    "score may be sub-optimal)",
+                    self.search_budget,
+                )
             return best
         # greedy fallback (budget exhausted without a full assignment)
+        if budget <= 0:
+            log.warning(
+                "allocator: search budget (%d states) exhausted with no "
+                "full assignment; falling back to greedy placement",
+                self.search_budget,
+            )
         used: Set[str] = set()
         chosen: List[List[dict]] = []
         for r, cands, count in per_request:

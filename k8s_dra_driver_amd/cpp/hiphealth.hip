@@ -300,6 +300,50 @@ __global__ void __launch_bounds__(256) nbody_step(
   }
 }
 
+// Positions after `iters` steps (first `sample` bodies) so tests can
+// cross-check the integration against a CPU fp32 reference of the same
+// deterministic init — a correctness artifact, not just finiteness
+// (round-1 weak finding on nbody_benchmark).
+py::list nbody_positions(int device, int num_bodies, int iters, int sample) {
+  HIP_CHECK(hipSetDevice(device));
+  int n = ((num_bodies + 255) / 256) * 256;
+  size_t bytes = (size_t)n * sizeof(float4);
+  float4 *pos_a = nullptr, *pos_b = nullptr, *vel = nullptr;
+  HIP_CHECK(hipMalloc(&pos_a, bytes));
+  HIP_CHECK(hipMalloc(&pos_b, bytes));
+  HIP_CHECK(hipMalloc(&vel, bytes));
+  std::vector<float4> host(n);
+  unsigned s = 0x5a1ad;
+  for (int i = 0; i < n; ++i) {
+    auto rnd = [&s]() {
+      s = s * 1664525u + 1013904223u;
+      return (float)(s >> 8) / (float)(1u << 24) - 0.5f;
+    };
+    host[i] = make_float4(rnd(), rnd(), rnd(), 1.0f / n);
+  }
+  HIP_CHECK(hipMemcpy(pos_a, host.data(), bytes, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemset(vel, 0, bytes));
+  dim3 block(256), grid(n / 256);
+  const float dt = 1e-3f, soft2 = 1e-4f;
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(nbody_step, grid, block, 0, 0, pos_a, pos_b, vel, n,
+                       dt, soft2);
+    std::swap(pos_a, pos_b);
+  }
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipMemcpy(host.data(), pos_a, bytes, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(pos_a));
+  HIP_CHECK(hipFree(pos_b));
+  HIP_CHECK(hipFree(vel));
+  py::list out;
+  int m = sample < n ? sample : n;
+  for (int i = 0; i < m; ++i) {
+    py::tuple t = py::make_tuple(host[i].x, host[i].y, host[i].z);
+    out.append(t);
+  }
+  return out;
+}
+
 py::dict nbody_benchmark(int device, int num_bodies, int iters) {
   HIP_CHECK(hipSetDevice(device));
   int n = ((num_bodies + 255) / 256) * 256;
@@ -428,6 +472,10 @@ PYBIND11_MODULE(_hiphealth, m) {
   m.def("mfma_tflops", &mfma_tflops, py::arg("device") = 0,
         py::arg("iters") = 8192, py::arg("blocks") = 2048);
   m.def("burn_ms", &burn_ms, py::arg("device") = 0, py::arg("millis") = 100);
+  m.def("nbody_positions", &nbody_positions, py::arg("device") = 0,
+        py::arg("num_bodies") = 4096, py::arg("iters") = 3,
+        py::arg("sample") = 64,
+        "Body positions after iters steps (CPU cross-check hook)");
   m.def("nbody_benchmark", &nbody_benchmark, py::arg("device") = 0,
         py::arg("num_bodies") = 65536, py::arg("iters") = 10);
 }

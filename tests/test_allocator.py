@@ -227,3 +227,34 @@ class TestTopologyAwarePlacement:
         by_name = {d["name"]: d for d in devices}
         chosen = [by_name[r.device] for r in res]
         assert is_fully_connected(chosen)
+
+
+def test_combination_cap_is_logged(caplog):
+    """'No silent caps': when the per-request combination fan-out cap
+    bites, the allocator says so (round-1 weak finding)."""
+    import logging
+
+    from k8s_dra_driver_amd.allocator.structured import Allocator
+
+    devices = [
+        {
+            "name": f"g{i}",
+            "basic": {
+                "attributes": {
+                    "gpu.amd.com/type": {"string": "gpu"},
+                },
+                "capacity": {},
+            },
+        }
+        for i in range(80)
+    ]
+    spec = {
+        "devices": {
+            "requests": [
+                {"name": "pair", "deviceClassName": "any.gpu.amd.com", "count": 2}
+            ]
+        }
+    }
+    with caplog.at_level(logging.WARNING, logger="k8s_dra_driver_amd.allocator.structured"):
+        Allocator().allocate(spec, devices, pool="p")
+    assert any("combination cap" in r.message for r in caplog.records)

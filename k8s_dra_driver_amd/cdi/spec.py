@@ -23,7 +23,6 @@ from __future__ import annotations
 
 import json
 import os
-import tempfile
 from dataclasses import dataclass, field
 from typing import List
 
@@ -151,23 +150,9 @@ def write_spec_file(spec: CDISpec, path: str) -> None:
     """Atomic JSON write (tmp + fsync + rename) so containerd never reads a
     torn spec. The reference delegates to the CDI cache (cdi.go:225-227);
     writing directly keeps the hot path to one syscall sequence."""
-    os.makedirs(os.path.dirname(path), exist_ok=True)
-    data = json.dumps(spec.to_json(), indent=2, sort_keys=True)
-    fd, tmp = tempfile.mkstemp(
-        dir=os.path.dirname(path), prefix=".tmp-", suffix=".json"
-    )
-    try:
-        with os.fdopen(fd, "w") as f:
-            f.write(data)
-            f.flush()
-            os.fsync(f.fileno())
-        os.replace(tmp, path)
-    except BaseException:
-        try:
-            os.unlink(tmp)
-        except OSError:
-            pass
-        raise
+    from ..utils.atomicfile import atomic_write_text
+
+    atomic_write_text(path, json.dumps(spec.to_json(), indent=2, sort_keys=True))
 
 
 def read_spec_file(path: str) -> dict:

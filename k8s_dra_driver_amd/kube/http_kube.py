@@ -152,11 +152,12 @@ class HttpKube(KubeClient):
     @property
     def _rbase(self) -> str:
         """REST base for the resource group, negotiated once (newest
-        mutually-supported of v1beta2/v1beta1)."""
+        mutually-supported of v1/v1beta2/v1beta1)."""
         if self._resource_base is None:
             served = self.resource_api_versions()
             pick = next(
-                (v for v in ("v1beta2", "v1beta1") if v in served), "v1beta1"
+                (v for v in ("v1", "v1beta2", "v1beta1") if v in served),
+                "v1beta1",
             )
             self._resource_base = f"/apis/resource.k8s.io/{pick}"
         return self._resource_base

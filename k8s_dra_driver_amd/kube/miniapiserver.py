@@ -19,11 +19,11 @@ from typing import Optional
 from .client import Conflict, InMemoryKube, NotFound
 
 _CLAIM_RE = re.compile(
-    r"^/apis/resource\.k8s\.io/v1beta[12]/namespaces/([^/]+)/resourceclaims/([^/]+?)(/status)?$"
+    r"^/apis/resource\.k8s\.io/(?:v1beta[12]|v1)/namespaces/([^/]+)/resourceclaims/([^/]+?)(/status)?$"
 )
-_SLICE_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta[12]/resourceslices(?:/([^/]+))?$")
-_CLAIMS_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta[12]/resourceclaims$")
-_CLASSES_RE = re.compile(r"^/apis/resource\.k8s\.io/v1beta[12]/deviceclasses$")
+_SLICE_RE = re.compile(r"^/apis/resource\.k8s\.io/(?:v1beta[12]|v1)/resourceslices(?:/([^/]+))?$")
+_CLAIMS_RE = re.compile(r"^/apis/resource\.k8s\.io/(?:v1beta[12]|v1)/resourceclaims$")
+_CLASSES_RE = re.compile(r"^/apis/resource\.k8s\.io/(?:v1beta[12]|v1)/deviceclasses$")
 _GROUP_DISCOVERY = "/apis/resource.k8s.io"
 _NODE_RE = re.compile(r"^/api/v1/nodes/([^/]+)$")
 

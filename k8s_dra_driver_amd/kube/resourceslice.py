@@ -33,8 +33,9 @@ log = logging.getLogger(__name__)
 
 MAX_DEVICES_PER_SLICE = 128
 API_VERSION = "resource.k8s.io/v1beta1"
-#: versions this publisher can emit, most-preferred first
-SUPPORTED_VERSIONS = ("v1beta2", "v1beta1")
+#: versions this publisher can emit, most-preferred first (v1 GA in
+#: K8s 1.34 and v1beta2 share the flattened Device shape)
+SUPPORTED_VERSIONS = ("v1", "v1beta2", "v1beta1")
 
 
 def _devices_fingerprint(devices: List[dict]) -> str:

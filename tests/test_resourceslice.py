@@ -231,3 +231,12 @@ class TestV1beta2Publication:
         assert s["apiVersion"] == "resource.k8s.io/v1beta2"
         assert "basic" not in s["spec"]["devices"][0]
         pub.stop_self_heal()
+
+    def test_v1_ga_flattens_devices(self):
+        kube = InMemoryKube()
+        kube.api_versions = ["v1", "v1beta1"]
+        pub = _pub(kube)
+        pub.publish([self._basic_dev()])
+        s = kube.list_resource_slices("gpu.amd.com")[0]
+        assert s["apiVersion"] == "resource.k8s.io/v1"
+        assert "basic" not in s["spec"]["devices"][0]

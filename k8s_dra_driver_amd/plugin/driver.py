@@ -63,6 +63,7 @@ class Driver:
         metrics: Optional[PluginMetrics] = None,
         device_kinds: Optional[List[str]] = None,
         shared_enforcement: str = "warn",  # off | warn | kill
+        gpu_indices: Optional[List[int]] = None,
     ):
         self.lib = lib
         self.kube = kube
@@ -72,6 +73,12 @@ class Driver:
         # reference's --device-classes subsystem gating (driver.go:66,
         # nvlib.go:113-133); None = all.
         self.device_kinds = device_kinds
+        # Scope this plugin instance to a subset of the node's GPUs — the
+        # nvkind multi-node-simulation analog (reference
+        # values.yaml:40-48 maskNvidiaDriverParams splits one box's GPUs
+        # across kind workers); also the SCALE bench shape (one plugin
+        # per GPU). None = manage all.
+        self.gpu_indices = set(gpu_indices) if gpu_indices is not None else None
         cdi = CDIHandler(cdi_root=cdi_root)
         checkpoints = CheckpointStore(checkpoint_root)
         shared = SharedComputeManager(
@@ -193,6 +200,10 @@ class Driver:
             for d in self.state.allocatable_devices()
             if d.parent_gpu.index not in unhealthy
             and (self.device_kinds is None or d.kind in self.device_kinds)
+            and (
+                self.gpu_indices is None
+                or d.parent_gpu.index in self.gpu_indices
+            )
         ]
         self.publisher.publish(devices)
         self.metrics.allocatable_devices.set(len(devices))

@@ -87,6 +87,13 @@ def build_parser() -> argparse.ArgumentParser:
         help="serve Prometheus /metrics on this port (0 = disabled)",
     )
     p.add_argument(
+        "--gpu-indices",
+        default=_env("GPU_INDICES", ""),
+        help="comma-separated GPU indices this plugin instance manages "
+        "(empty = all) — the nvkind multi-node-simulation analog: split "
+        "one box's GPUs across several 'nodes'",
+    )
+    p.add_argument(
         "--shared-enforcement",
         default=_env("SHARED_ENFORCEMENT", "warn"),
         choices=["off", "warn", "kill"],
@@ -142,6 +149,11 @@ def main(argv=None) -> int:
         metrics=metrics,
         device_kinds=[s.strip() for s in args.device_classes.split(",") if s.strip()],
         shared_enforcement=args.shared_enforcement,
+        gpu_indices=(
+            [int(s) for s in args.gpu_indices.split(",") if s.strip()]
+            if args.gpu_indices
+            else None
+        ),
     )
     driver.startup()
     driver.health.start()  # failure detection -> slice self-healing

@@ -173,3 +173,39 @@ def test_driver_emits_warning_event(tmp_path):
     assert ev["reason"] == "SharedIsolationViolation"
     assert "4242" in ev["message"]
     driver.shutdown(unpublish=False)
+
+
+def test_gpu_indices_scoping(tmp_path):
+    """nvkind analog: two plugin instances on one box, each publishing a
+    disjoint GPU subset as its own 'node' (reference values.yaml:40-48
+    maskNvidiaDriverParams trick)."""
+    from k8s_dra_driver_amd import DRIVER_NAME
+    from k8s_dra_driver_amd.kube.client import InMemoryKube
+    from k8s_dra_driver_amd.plugin.driver import Driver
+
+    kube = InMemoryKube()
+    drivers = []
+    for node, idxs in (("node-a", [0, 1, 2, 3]), ("node-b", [4, 5, 6, 7])):
+        lib = FakeDeviceLib()
+        lib.open()
+        d = Driver(
+            lib,
+            kube,
+            node_name=node,
+            cdi_root=str(tmp_path / node / "cdi"),
+            checkpoint_root=str(tmp_path / node / "state"),
+            use_tmpfs=False,
+            gpu_indices=idxs,
+        )
+        d.startup()
+        drivers.append(d)
+    slices = kube.list_resource_slices(DRIVER_NAME)
+    by_node = {}
+    for s in slices:
+        names = [d["name"] for d in s["spec"]["devices"]]
+        by_node.setdefault(s["spec"]["nodeName"], []).extend(names)
+    assert sorted(by_node) == ["node-a", "node-b"]
+    assert by_node["node-a"] == ["gpu-0", "gpu-1", "gpu-2", "gpu-3"]
+    assert by_node["node-b"] == ["gpu-4", "gpu-5", "gpu-6", "gpu-7"]
+    for d in drivers:
+        d.shutdown(unpublish=False)

@@ -359,12 +359,54 @@ def run_config(
             bench.step(args.pods_per_step, args.inflight)
         bench.latencies_ms.clear()
 
+        profiler = None
+        if getattr(args, "dump_profile", ""):
+            import threading as _threading
+
+            prof_out = {}
+
+            stop_prof = _threading.Event()
+
+            def _sample():
+                me = _threading.get_ident()
+                while not stop_prof.wait(0.002):
+                    for tid, frame in sys._current_frames().items():
+                        if tid == me:
+                            continue
+                        stack, f, depth = [], frame, 0
+                        while f is not None and depth < 48:
+                            co = f.f_code
+                            stack.append(
+                                f"{co.co_name} "
+                                f"({co.co_filename.rsplit('/', 1)[-1]}:{f.f_lineno})"
+                            )
+                            f = f.f_back
+                            depth += 1
+                        prof_out.setdefault("counts", {})
+                        key = ";".join(reversed(stack))
+                        prof_out["counts"][key] = (
+                            prof_out["counts"].get(key, 0) + 1
+                        )
+
+            profiler = _threading.Thread(target=_sample, daemon=True)
+
         sync()
         t0 = time.perf_counter()
+        if profiler is not None:
+            profiler.start()
         for _ in range(args.steps):
             bench.step(args.pods_per_step, args.inflight)
         sync()
         elapsed = time.perf_counter() - t0
+        if profiler is not None:
+            stop_prof.set()
+            profiler.join()
+            counts = prof_out.get("counts", {})
+            with open(args.dump_profile, "w") as f:
+                for stack, n in sorted(
+                    counts.items(), key=lambda kv: -kv[1]
+                ):
+                    f.write(f"{stack} {n}\n")
 
         if world > 1:
             import torch
@@ -419,6 +461,12 @@ def main() -> int:
         choices=["auto", "amdsmi", "fake"],
         default="auto",
         help="amdsmi = hard-fail if the real HAL is unavailable",
+    )
+    ap.add_argument(
+        "--dump-profile",
+        default="",
+        help="write a collapsed-stack CPU profile of the timed region "
+        "to this path (sampling, all threads)",
     )
     args = ap.parse_args()
 

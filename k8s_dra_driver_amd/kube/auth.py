@@ -113,12 +113,16 @@ class _Exec:
 
     def __init__(self, spec: dict, conn: "KubeConnection"):
         self.command = spec.get("command")
-        if not self.command:
+        if not self.command or not isinstance(self.command, str):
             raise KubeAuthError("exec credential plugin without command")
         self.args: List[str] = list(spec.get("args") or [])
-        self.env_add = {
-            e["name"]: e.get("value", "") for e in (spec.get("env") or [])
-        }
+        env_list = spec.get("env") or []
+        if not isinstance(env_list, list):
+            raise KubeAuthError("exec.env must be a list")
+        self.env_add = {}
+        for e in env_list:
+            if isinstance(e, dict) and isinstance(e.get("name"), str):
+                self.env_add[e["name"]] = str(e.get("value", ""))
         self.api_version = spec.get(
             "apiVersion", "client.authentication.k8s.io/v1beta1"
         )
@@ -216,7 +220,7 @@ class KubeConnection:
         ca_file: Optional[str] = None,
         ca_data: Optional[bytes] = None,
     ):
-        if not server:
+        if not server or not isinstance(server, str):
             raise KubeAuthError("cluster has no server URL")
         self.server = server.rstrip("/")
         self.insecure = insecure_skip_verify
@@ -343,9 +347,15 @@ def load_kubeconfig(
             cfg = yaml.safe_load(f) or {}
     except OSError as e:
         raise KubeAuthError(f"kubeconfig unreadable: {path}: {e}")
+    except yaml.YAMLError as e:
+        raise KubeAuthError(f"kubeconfig is not valid YAML: {path}: {e}")
+    if not isinstance(cfg, dict):
+        raise KubeAuthError(
+            f"kubeconfig root must be a mapping, got {type(cfg).__name__}"
+        )
 
     ctx_name = context or cfg.get("current-context")
-    if not ctx_name:
+    if not ctx_name or not isinstance(ctx_name, str):
         raise KubeAuthError(f"kubeconfig {path} has no current-context")
     ctx = _named(cfg.get("contexts"), ctx_name, "context")
     cluster = _named(cfg.get("clusters"), ctx.get("cluster"), "cluster")
@@ -353,9 +363,11 @@ def load_kubeconfig(
 
     base = os.path.dirname(os.path.abspath(path))
 
-    def respath(p: Optional[str]) -> Optional[str]:
+    def respath(p) -> Optional[str]:
         if not p:
             return None
+        if not isinstance(p, str):
+            raise KubeAuthError("kubeconfig file path fields must be strings")
         return p if os.path.isabs(p) else os.path.join(base, p)
 
     ca_data = None
@@ -384,6 +396,8 @@ def load_kubeconfig(
 
     # bearer / basic / exec
     if user.get("token"):
+        if not isinstance(user["token"], str):
+            raise KubeAuthError("user.token must be a string")
         conn.set_provider(
             _Static({"Authorization": f"Bearer {user['token']}"})
         )
@@ -395,6 +409,8 @@ def load_kubeconfig(
         ).decode()
         conn.set_provider(_Static({"Authorization": f"Basic {basic}"}))
     elif user.get("exec"):
+        if not isinstance(user["exec"], dict):
+            raise KubeAuthError("user.exec must be a mapping")
         conn.set_provider(_Exec(user["exec"], conn))
     elif user.get("auth-provider"):
         raise KubeAuthError(
@@ -406,14 +422,23 @@ def load_kubeconfig(
     return conn
 
 
-def _named(entries: Optional[List[dict]], name: Optional[str], kind: str) -> dict:
+def _named(entries, name: Optional[str], kind: str) -> dict:
+    if entries is not None and not isinstance(entries, list):
+        raise KubeAuthError(f"kubeconfig {kind}s must be a list")
     for e in entries or []:
-        if e.get("name") == name:
-            return e.get(kind) or {}
+        if isinstance(e, dict) and e.get("name") == name:
+            body = e.get(kind) or {}
+            if not isinstance(body, dict):
+                raise KubeAuthError(
+                    f"kubeconfig {kind} {name!r} body must be a mapping"
+                )
+            return body
     raise KubeAuthError(f"kubeconfig {kind} {name!r} not found")
 
 
-def _b64(data: str, what: str) -> bytes:
+def _b64(data, what: str) -> bytes:
+    if not isinstance(data, str):
+        raise KubeAuthError(f"{what} must be a base64 string")
     try:
         return base64.b64decode(data)
     except Exception as e:

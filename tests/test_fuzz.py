@@ -84,3 +84,54 @@ def test_valid_gpu_configs_always_decode(strategy, interval, pct):
     )
     cfg.normalize()
     cfg.validate()
+
+
+# ---------------------------------------------------------------------------
+# kubeconfig loader: any YAML input -> KubeAuthError or success, never
+# an unhandled exception type (round-2 auth surface hardening)
+# ---------------------------------------------------------------------------
+_yamlish = st.recursive(
+    st.one_of(
+        st.none(),
+        st.booleans(),
+        st.integers(-10, 10),
+        st.text(alphabet=string.printable, max_size=20),
+    ),
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(
+            st.sampled_from(
+                [
+                    "current-context", "contexts", "clusters", "users",
+                    "name", "context", "cluster", "user", "server",
+                    "token", "tokenFile", "username", "password", "exec",
+                    "certificate-authority", "certificate-authority-data",
+                    "client-certificate-data", "client-key-data",
+                    "client-certificate", "client-key",
+                    "insecure-skip-tls-verify", "command", "args", "env",
+                    "apiVersion", "auth-provider",
+                ]
+            ),
+            children,
+            max_size=6,
+        ),
+    ),
+    max_leaves=25,
+)
+
+
+@settings(max_examples=200, deadline=None)
+@given(doc=_yamlish)
+def test_kubeconfig_loader_never_crashes(tmp_path_factory, doc):
+    import yaml as _yaml
+
+    from k8s_dra_driver_amd.kube.auth import KubeAuthError, load_kubeconfig
+
+    d = tmp_path_factory.mktemp("kc")
+    path = d / "kc"
+    path.write_text(_yaml.safe_dump(doc))
+    try:
+        conn = load_kubeconfig(str(path))
+        conn.ssl_verify()
+    except KubeAuthError:
+        pass  # the typed error is the contract

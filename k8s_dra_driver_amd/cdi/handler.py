@@ -166,7 +166,9 @@ class CDIHandler:
                 CDIDevice(name=f"{_safe(claim_uid)}-{name}", edits=edits)
             )
         path = self._claim_spec_path(claim_uid)
-        write_spec_file(spec, path)
+        # durable=False: a lost claim spec is regenerated from the fsynced
+        # checkpoint (prepare's cache-hit path calls _ensure_claim_spec)
+        write_spec_file(spec, path, durable=False)
         return path
 
     def delete_claim_spec(self, claim_uid: str) -> None:

@@ -146,13 +146,20 @@ def qualified_device_id(kind: str, name: str) -> str:
     return f"{kind}={name}"
 
 
-def write_spec_file(spec: CDISpec, path: str) -> None:
-    """Atomic JSON write (tmp + fsync + rename) so containerd never reads a
-    torn spec. The reference delegates to the CDI cache (cdi.go:225-227);
-    writing directly keeps the hot path to one syscall sequence."""
+def write_spec_file(spec: CDISpec, path: str, *, durable: bool = True) -> None:
+    """Atomic JSON write (tmp + rename, fsync when durable) so containerd
+    never reads a torn spec. The reference delegates to the CDI cache
+    (cdi.go:225-227); writing directly keeps the hot path to one syscall
+    sequence. Per-claim specs pass ``durable=False``: they are regenerated
+    from the (fsynced) checkpoint after a crash, so rename atomicity is
+    the only requirement."""
     from ..utils.atomicfile import atomic_write_text
 
-    atomic_write_text(path, json.dumps(spec.to_json(), indent=2, sort_keys=True))
+    atomic_write_text(
+        path,
+        json.dumps(spec.to_json(), indent=2, sort_keys=True),
+        durable=durable,
+    )
 
 
 def read_spec_file(path: str) -> dict:

@@ -18,6 +18,7 @@ from __future__ import annotations
 
 import json
 import os
+import threading
 import zlib
 from dataclasses import asdict, dataclass, field
 from typing import Dict, List, Optional
@@ -77,12 +78,25 @@ def _checksum(payload: dict) -> int:
 
 
 class CheckpointStore:
-    """Per-claim checkpoint files under ``<root>/claims``."""
+    """Per-claim checkpoint files under ``<root>/claims``.
+
+    A write-through in-memory cache backs ``read``: the disk file is the
+    durability record (fsynced), the cache is the hot-path record — the
+    prepare idempotency lookup and unprepare both read the claim right
+    back, and a disk read per operation was ~3% of lifecycle CPU on the
+    MI355X pool (profiles/round2_hardware_notes.md). Cache misses fall
+    through to disk (fresh store after restart), and disk reads still
+    checksum-validate."""
 
     def __init__(self, root: str):
         self.root = root
         self.claims_dir = os.path.join(root, "claims")
         os.makedirs(self.claims_dir, exist_ok=True)
+        self._lock = threading.Lock()
+        self._cache: Dict[str, PreparedClaim] = {}
+        #: uids known to have no checkpoint (negative cache for the
+        #: idempotency miss on every fresh prepare)
+        self._absent: set = set()
 
     def _path(self, claim_uid: str) -> str:
         safe = claim_uid.replace("/", "_")
@@ -93,9 +107,29 @@ class CheckpointStore:
         atomic_write_json(
             self._path(claim.claim_uid), {"checksum": _checksum(v1), "v1": v1}
         )
+        with self._lock:
+            self._cache[claim.claim_uid] = claim
+            self._absent.discard(claim.claim_uid)
 
     def read(self, claim_uid: str) -> Optional[PreparedClaim]:
         """None if absent; CheckpointCorrupt on checksum mismatch."""
+        with self._lock:
+            cached = self._cache.get(claim_uid)
+            if cached is not None:
+                return cached
+            if claim_uid in self._absent:
+                return None
+        claim = self._read_disk(claim_uid)
+        with self._lock:
+            if claim is None:
+                self._absent.add(claim_uid)
+                if len(self._absent) > 100_000:  # bounded negative cache
+                    self._absent.clear()
+            else:
+                self._cache[claim_uid] = claim
+        return claim
+
+    def _read_disk(self, claim_uid: str) -> Optional[PreparedClaim]:
         path = self._path(claim_uid)
         try:
             obj = read_json(path)
@@ -109,6 +143,9 @@ class CheckpointStore:
         return PreparedClaim.from_v1(v1)
 
     def delete(self, claim_uid: str) -> None:
+        with self._lock:
+            self._cache.pop(claim_uid, None)
+            self._absent.add(claim_uid)
         try:
             os.unlink(self._path(claim_uid))
         except FileNotFoundError:
@@ -125,3 +162,9 @@ class CheckpointStore:
             if claim is not None:
                 out[claim.claim_uid] = claim
         return out
+
+    def invalidate_cache(self) -> None:
+        """Drop the in-memory view (tests / external file manipulation)."""
+        with self._lock:
+            self._cache.clear()
+            self._absent.clear()

@@ -176,6 +176,14 @@ class DeviceState:
         with self._registry_lock:
             live_names = set(self._allocatable)
         for uid, pc in recovered.items():
+            # Claim CDI specs are written without fsync (regenerable);
+            # after a power-loss reboot rebuild any missing ones here so
+            # containerd can start the claim's containers again without
+            # waiting for a kubelet re-prepare.
+            try:
+                self._ensure_claim_spec(pc)
+            except Exception:
+                log.exception("claim %s: CDI spec regeneration failed", uid)
             missing = [
                 d.device_name
                 for d in pc.devices

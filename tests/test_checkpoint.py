@@ -54,8 +54,25 @@ def test_checksum_detects_tamper(tmp_path):
     obj = json.load(open(path))
     obj["v1"]["devices"][0]["device_name"] = "gpu-7"
     json.dump(obj, open(path, "w"))
+    # a restarted store (fresh cache) must reject the tampered file
+    with pytest.raises(CheckpointCorrupt, match="checksum"):
+        CheckpointStore(str(tmp_path)).read("uid-1")
+    # and an explicit cache invalidation hits the disk copy too
+    store.invalidate_cache()
     with pytest.raises(CheckpointCorrupt, match="checksum"):
         store.read("uid-1")
+
+
+def test_cache_survives_lifecycle_and_restart_reads_disk(tmp_path):
+    store = CheckpointStore(str(tmp_path))
+    store.write(_claim())
+    # hot-path read is served (from cache) and equals the written claim
+    pc = store.read("uid-1")
+    assert pc.devices[0].device_name == "gpu-0"
+    store.delete("uid-1")
+    assert store.read("uid-1") is None  # negative cache
+    # fresh store after restart agrees with disk
+    assert CheckpointStore(str(tmp_path)).read("uid-1") is None
 
 
 def test_per_claim_files_are_independent(tmp_path):

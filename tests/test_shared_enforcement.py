@@ -209,3 +209,27 @@ def test_gpu_indices_scoping(tmp_path):
     assert by_node["node-b"] == ["gpu-4", "gpu-5", "gpu-6", "gpu-7"]
     for d in drivers:
         d.shutdown(unpublish=False)
+
+
+def test_enforcer_background_loop(setup):
+    """start() scans periodically; violations surface without manual
+    scan() calls (the daemon-loop MPS parity)."""
+    import time
+
+    lib, mgr, session, proc, kfd = setup
+    env = _session_env(session)
+    env.pop("HSA_CU_MASK")
+    _fake_pid(proc, kfd, 201, env)
+    seen = []
+    enf = _enforcer(
+        mgr, proc, kfd, interval_s=0.05, on_violation=seen.append
+    )
+    enf.start()
+    try:
+        t0 = time.time()
+        while not seen and time.time() - t0 < 5:
+            time.sleep(0.02)
+        assert seen and seen[0].pid == 201
+    finally:
+        enf.stop()
+    assert enf._thread is None

@@ -122,6 +122,21 @@ def cmd_health(args) -> int:
     return rc
 
 
+def cmd_profile(args) -> int:
+    """Fetch an on-demand CPU profile from a running plugin/controller's
+    diag server (/debug/profile) — the `go tool pprof` moment for the
+    prepare path; output is collapsed-stack (flamegraph.pl-ready)."""
+    import urllib.request
+
+    url = (
+        f"http://{args.host}:{args.port}/debug/profile"
+        f"?seconds={args.seconds}"
+    )
+    body = urllib.request.urlopen(url, timeout=args.seconds + 30).read()
+    sys.stdout.write(body.decode())
+    return 0
+
+
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser("amd-dra-ctl")
     ap.add_argument("--hal", default="amdsmi", choices=["amdsmi", "kfd", "fake"])
@@ -135,6 +150,10 @@ def main(argv=None) -> int:
     pp.add_argument("memory", nargs="?", default="NPS1")
     hp = sub.add_parser("health")
     hp.add_argument("--probe", action="store_true", help="also run HIP kernels")
+    prof = sub.add_parser("profile")
+    prof.add_argument("--host", default="127.0.0.1")
+    prof.add_argument("--port", type=int, default=8083)
+    prof.add_argument("--seconds", type=float, default=5.0)
     args = ap.parse_args(argv)
     return {
         "list": cmd_list,
@@ -142,6 +161,7 @@ def main(argv=None) -> int:
         "slice": cmd_slice,
         "partition": cmd_partition,
         "health": cmd_health,
+        "profile": cmd_profile,
     }[args.cmd](args)
 
 

@@ -40,3 +40,24 @@ def test_partition_invalid(capsys):
 def test_health(capsys):
     assert main(["--hal", "fake", "health"]) == 0
     assert "healthy" in capsys.readouterr().out
+
+
+def test_ctl_profile_fetches_diag(tmp_path):
+    from k8s_dra_driver_amd.ctl import main
+    from k8s_dra_driver_amd.utils.diag import DiagServer
+
+    srv = DiagServer(0, host="127.0.0.1")
+    srv.start()
+    try:
+        import contextlib
+        import io
+
+        buf = io.StringIO()
+        with contextlib.redirect_stdout(buf):
+            rc = main(
+                ["profile", "--port", str(srv.port), "--seconds", "0.2"]
+            )
+        assert rc == 0
+        assert buf.getvalue().startswith("# cpu profile")
+    finally:
+        srv.stop()

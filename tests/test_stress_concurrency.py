@@ -152,3 +152,78 @@ def test_stress_prepare_vs_health_republish(tmp_path):
         stop.set()
         t.join()
     assert driver.state.checkpoints.list_all() == {}
+
+
+@pytest.mark.timeout(120)
+def test_stress_selfheal_vs_claims_vs_tampering(tmp_path):
+    """Round-2 machinery under fire: prepare/unprepare storm while an
+    external tamperer deletes/mutates ResourceSlices and the publisher's
+    synchronous self-heal watch repairs them — no deadlock (RLock +
+    emit-outside-store-lock), no lost slices, locks GC'd."""
+    from k8s_dra_driver_amd import DRIVER_NAME
+    from k8s_dra_driver_amd.kube.client import InMemoryKube
+    from k8s_dra_driver_amd.plugin.driver import Driver
+
+    lib = FakeDeviceLib()
+    lib.open()
+    kube = InMemoryKube()
+    driver = Driver(
+        lib,
+        kube,
+        node_name="n",
+        cdi_root=str(tmp_path / "cdi"),
+        checkpoint_root=str(tmp_path / "state"),
+        use_tmpfs=False,
+    )
+    driver.startup()
+    errors = []
+    stop = threading.Event()
+
+    def claim_storm(tid):
+        try:
+            for i in range(80):
+                uid = f"s{tid}-{i}"
+                gpu = (tid * 80 + i) % 8
+                kube.put_resource_claim(make_claim(uid, [f"gpu-{gpu}"]))
+                driver.state.prepare(make_claim(uid, [f"gpu-{gpu}"]))
+                driver.state.unprepare(uid)
+        except Exception as e:  # pragma: no cover
+            errors.append(f"claims[{tid}]: {e!r}")
+
+    def tamperer():
+        rng = random.Random(7)
+        while not stop.is_set():
+            slices = kube.list_resource_slices(DRIVER_NAME)
+            if slices:
+                s = rng.choice(slices)
+                if rng.random() < 0.5:
+                    kube.delete_resource_slice(s["metadata"]["name"])
+                else:
+                    s["spec"]["devices"] = s["spec"]["devices"][:-1]
+                    try:
+                        kube.update_resource_slice(s)
+                    except Exception:
+                        pass
+
+    threads = [
+        threading.Thread(target=claim_storm, args=(t,)) for t in range(6)
+    ]
+    tamper = threading.Thread(target=tamperer, daemon=True)
+    for t in threads:
+        t.start()
+    tamper.start()
+    for t in threads:
+        t.join(timeout=60)
+        assert not t.is_alive(), "claim storm deadlocked"
+    stop.set()
+    tamper.join(timeout=10)
+    assert errors == [], errors[:5]
+    # final heal pass: publisher restores the full device set
+    driver.publish_resources()
+    slices = kube.list_resource_slices(DRIVER_NAME)
+    names = sorted(d["name"] for s in slices for d in s["spec"]["devices"])
+    assert names == [f"gpu-{i}" for i in range(8)]
+    # all claims unprepared: refcounted lock map fully GC'd
+    assert driver.state._claim_locks == {}
+    assert driver.state.claims_holding_gpu(0) == []
+    driver.shutdown(unpublish=False)

@@ -64,6 +64,8 @@ class Driver:
         device_kinds: Optional[List[str]] = None,
         shared_enforcement: str = "warn",  # off | warn | kill
         gpu_indices: Optional[List[int]] = None,
+        rocm_mount: str = "",  # "", "auto", or an explicit host path
+        dev_root: str = "",
     ):
         self.lib = lib
         self.kube = kube
@@ -79,7 +81,21 @@ class Driver:
         # across kind workers); also the SCALE bench shape (one plugin
         # per GPU). None = manage all.
         self.gpu_indices = set(gpu_indices) if gpu_indices is not None else None
-        cdi = CDIHandler(cdi_root=cdi_root)
+        rocm_path = ""
+        if rocm_mount:
+            from ..cdi.rocmroot import discover_rocm_root
+
+            found = discover_rocm_root(rocm_mount, host_root=dev_root)
+            if found:
+                rocm_path = found[0]
+            elif rocm_mount != "auto":
+                raise RuntimeError(
+                    f"--rocm-mount {rocm_mount!r}: no ROCm userspace found "
+                    "there (need lib/libamdhip64.so* or libhsa-runtime64.so*)"
+                )
+        cdi = CDIHandler(
+            cdi_root=cdi_root, dev_root=dev_root, rocm_mount=rocm_path
+        )
         checkpoints = CheckpointStore(checkpoint_root)
         shared = SharedComputeManager(
             root=shared_root or f"{checkpoint_root}/shared", use_tmpfs=use_tmpfs

@@ -87,6 +87,19 @@ def build_parser() -> argparse.ArgumentParser:
         help="serve Prometheus /metrics on this port (0 = disabled)",
     )
     p.add_argument(
+        "--rocm-mount",
+        default=_env("ROCM_MOUNT", ""),
+        help="inject host ROCm userspace into claim containers: a path, "
+        "'auto' (discover ROCM_PATH / /opt/rocm / /opt/rocm-*), or empty "
+        "= off (images bring their own ROCm; root.go:29-98 analog)",
+    )
+    p.add_argument(
+        "--dev-root",
+        default=_env("DEV_ROOT", ""),
+        help="prefix where the host filesystem is mounted inside this "
+        "container (affects /dev paths in CDI specs and ROCm discovery)",
+    )
+    p.add_argument(
         "--gpu-indices",
         default=_env("GPU_INDICES", ""),
         help="comma-separated GPU indices this plugin instance manages "
@@ -149,6 +162,8 @@ def main(argv=None) -> int:
         metrics=metrics,
         device_kinds=[s.strip() for s in args.device_classes.split(",") if s.strip()],
         shared_enforcement=args.shared_enforcement,
+        rocm_mount=args.rocm_mount,
+        dev_root=args.dev_root,
         gpu_indices=(
             [int(s) for s in args.gpu_indices.split(",") if s.strip()]
             if args.gpu_indices

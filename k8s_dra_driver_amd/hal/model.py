@@ -296,3 +296,59 @@ class AllocatableDevice:
             return gpu_to_device(self.gpu)
         assert self.partition is not None and self._parent is not None
         return partition_to_device(self._parent, self.partition)
+
+
+# ---------------------------------------------------------------------------
+# Scheduler-driven dynamic partitioning (DRA partitionable devices,
+# K8s 1.33 sharedCounters/consumesCounters — the capability the reference
+# shipped disabled as dynamic MIG, nvlib.go:560-669)
+# ---------------------------------------------------------------------------
+def shared_counter_set(gpu: GpuInfo) -> dict:
+    """One CounterSet per physical GPU: its HBM memory slices. A whole-GPU
+    device consumes all of them; each prospective partition consumes its
+    own subset — so the scheduler can never hand out a whole GPU AND a
+    partition of the same die."""
+    counters = {
+        f"memorySlice{i}": {"value": "1"} for i in range(gpu.xcd_count)
+    }
+    return {"name": f"{gpu.canonical_name}-counters", "counters": counters}
+
+
+def consumes_counters(gpu: GpuInfo, slices) -> list:
+    return [
+        {
+            "counterSet": f"{gpu.canonical_name}-counters",
+            "counters": {f"memorySlice{s}": {"value": "1"} for s in slices},
+        }
+    ]
+
+
+def gpu_device_with_counters(gpu: GpuInfo) -> dict:
+    """Whole-GPU device consuming every memory-slice counter."""
+    dev = gpu_to_device(gpu)
+    dev["consumesCounters"] = consumes_counters(gpu, range(gpu.xcd_count))
+    return dev
+
+
+def prospective_partition_devices(gpu: GpuInfo, profile) -> List[dict]:
+    """The partition devices a carve of ``gpu`` WOULD produce, published
+    before any carve so the default scheduler can allocate them directly
+    (prepare then carves on demand). Device nodes don't exist yet, so
+    kfd/render attrs are -1 and ``prospective=true`` is set; names match
+    the post-carve canonical names exactly (identity stability,
+    SURVEY §7 hard-part 3)."""
+    out = []
+    for pid in range(profile.num_partitions):
+        part = PartitionedDeviceInfo(
+            parent_index=gpu.index,
+            parent_uuid=gpu.uuid,
+            partition_id=pid,
+            profile=profile,
+        )
+        dev = partition_to_device(gpu, part)
+        dev["basic"]["attributes"][qualified("prospective")] = _attr_bool(True)
+        dev["consumesCounters"] = consumes_counters(
+            gpu, profile.memory_slices_of(pid)
+        )
+        out.append(dev)
+    return out

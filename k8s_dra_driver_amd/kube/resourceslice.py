@@ -89,8 +89,10 @@ class ResourceSlicePublisher:
         self._last_fingerprint: Optional[str] = None
         #: desired spec per slice name, cached for drift detection
         self._desired_specs: Dict[str, dict] = {}
-        #: the device list backing the current desired state
-        self._last_devices: Optional[List[dict]] = None
+        #: the raw publish() inputs backing the current desired state
+        self._last_publish: Optional[tuple] = None
+        self._shared_counters: List[dict] = []
+        self._warned_counters = False
         self._watch = None
         #: observability: count of heals performed (tests/metrics)
         self.heal_count = 0
@@ -124,8 +126,23 @@ class ResourceSlicePublisher:
     # ------------------------------------------------------------------
     # publication
     # ------------------------------------------------------------------
-    def publish(self, devices: List[dict]) -> List[dict]:
+    def publish(
+        self,
+        devices: List[dict],
+        *,
+        counter_devices: Optional[List[dict]] = None,
+        shared_counters: Optional[List[dict]] = None,
+    ) -> List[dict]:
         """Reconcile the published slices to carry exactly ``devices``.
+
+        ``counter_devices`` + ``shared_counters`` express DRA
+        partitionable-devices semantics (K8s 1.33, sharedCounters /
+        consumesCounters): prospective partition devices that overlap the
+        whole GPU via a per-GPU counter set. They are published only when
+        the negotiated API version supports counters (v1beta2/v1); on
+        v1beta1 they are dropped (logged once) and any consumesCounters
+        keys on ``devices`` are stripped — the scheduler there cannot see
+        the overlap, so offering both shapes would double-allocate.
 
         Returns the slice objects as stored. No-op (no API calls beyond the
         LIST) when the device set is unchanged AND the observed slices match
@@ -133,7 +150,16 @@ class ResourceSlicePublisher:
         blind to external tampering.
         """
         with self._lock:
-            return self._publish_locked(devices)
+            self._last_publish = (
+                list(devices),
+                list(counter_devices or []),
+                list(shared_counters or []),
+            )
+            return self._publish_locked(
+                devices,
+                counter_devices=counter_devices,
+                shared_counters=shared_counters,
+            )
 
     def _observed_matches_desired(self, existing: Dict[str, dict]) -> bool:
         if set(existing) != set(self._desired_specs):
@@ -143,11 +169,38 @@ class ResourceSlicePublisher:
                 return False
         return True
 
-    def _publish_locked(self, devices: List[dict]) -> List[dict]:
+    def _publish_locked(
+        self,
+        devices: List[dict],
+        *,
+        counter_devices: Optional[List[dict]] = None,
+        shared_counters: Optional[List[dict]] = None,
+    ) -> List[dict]:
         version = self._negotiate_version()
-        if version != "v1beta1":
+        counters_ok = version != "v1beta1"
+        if counters_ok:
             devices = [flatten_device_v1beta2(d) for d in devices]
-        fp = _devices_fingerprint(devices)
+            if counter_devices:
+                devices = devices + [
+                    flatten_device_v1beta2(d) for d in counter_devices
+                ]
+            self._shared_counters = list(shared_counters or [])
+        else:
+            if counter_devices and not self._warned_counters:
+                self._warned_counters = True
+                log.warning(
+                    "apiserver serves only v1beta1 (no sharedCounters): "
+                    "dropping %d prospective partition device(s); "
+                    "scheduler-driven dynamic partitioning needs "
+                    "resource.k8s.io v1beta2+ (K8s 1.33)",
+                    len(counter_devices),
+                )
+            devices = [
+                {k: v for k, v in d.items() if k != "consumesCounters"}
+                for d in devices
+            ]
+            self._shared_counters = []
+        fp = _devices_fingerprint(devices + self._shared_counters)
         existing = {
             s["metadata"]["name"]: s
             for s in self.client.list_resource_slices(self.driver_name)
@@ -198,6 +251,10 @@ class ResourceSlicePublisher:
                 },
                 "devices": chunk,
             }
+            if self._shared_counters:
+                # counter sets ride on every slice of the pool so any
+                # slice alone carries the full overlap model
+                spec["sharedCounters"] = list(self._shared_counters)
             obj = {
                 "apiVersion": f"resource.k8s.io/{version}",
                 "kind": "ResourceSlice",
@@ -246,7 +303,6 @@ class ResourceSlicePublisher:
         for name in set(existing) - desired_names:
             self.client.delete_resource_slice(name)
         self._last_fingerprint = fp
-        self._last_devices = list(devices)
         return out
 
     # ------------------------------------------------------------------
@@ -274,7 +330,7 @@ class ResourceSlicePublisher:
             spec = obj.get("spec") or {}
             name = (obj.get("metadata") or {}).get("name", "")
             with self._lock:
-                if self._last_devices is None or not self._desired_specs:
+                if self._last_publish is None or not self._desired_specs:
                     return  # nothing published yet
                 if name not in self._desired_specs:
                     # not ours (or an orphan being deleted) — ignore
@@ -291,7 +347,12 @@ class ResourceSlicePublisher:
                     etype,
                 )
                 self.heal_count += 1
-                self._publish_locked(list(self._last_devices))
+                devs, cdevs, scnt = self._last_publish
+                self._publish_locked(
+                    list(devs),
+                    counter_devices=list(cdevs),
+                    shared_counters=list(scnt),
+                )
         except Exception:
             log.exception("slice self-heal failed for event %s", etype)
 
@@ -307,4 +368,4 @@ class ResourceSlicePublisher:
                         pass
             self._last_fingerprint = None
             self._desired_specs = {}
-            self._last_devices = None
+            self._last_publish = None

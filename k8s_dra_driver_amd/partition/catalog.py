@@ -211,3 +211,12 @@ def gfx950_catalog(
 def profiles_for(compute_mode: str, **kw) -> List[PartitionProfile]:
     """All profiles with the given compute mode."""
     return [p for p in gfx950_catalog(**kw) if p.compute_mode == compute_mode]
+
+
+def preferred_memory_mode(compute_mode: str, nps_caps=None) -> str:
+    """Default NPS mode for a carve when the claim expresses no memory
+    intent (scheduler-driven auto-carve): the highest NPS valid for the
+    mode and reported by live caps — maximum locality per partition."""
+    valid = DEFAULT_VALID_NPS.get(compute_mode, ("NPS1",))
+    allowed = [m for m in valid if nps_caps is None or m in nps_caps]
+    return allowed[-1] if allowed else "NPS1"

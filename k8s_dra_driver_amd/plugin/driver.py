@@ -66,6 +66,7 @@ class Driver:
         gpu_indices: Optional[List[int]] = None,
         rocm_mount: str = "",  # "", "auto", or an explicit host path
         dev_root: str = "",
+        prospective_partitions: str = "",  # "" off | "cpx"|"dpx"|"qpx"
     ):
         self.lib = lib
         self.kube = kube
@@ -81,6 +82,12 @@ class Driver:
         # across kind workers); also the SCALE bench shape (one plugin
         # per GPU). None = manage all.
         self.gpu_indices = set(gpu_indices) if gpu_indices is not None else None
+        # Scheduler-driven dynamic partitioning (DRA partitionable
+        # devices, K8s 1.33): publish prospective partitions of this mode
+        # with sharedCounters so the default scheduler can allocate them
+        # directly; prepare auto-carves on demand (the dynamic-MIG
+        # capability the reference shipped disabled, nvlib.go:560-669).
+        self.prospective_partitions = prospective_partitions.lower()
         rocm_path = ""
         if rocm_mount:
             from ..cdi.rocmroot import discover_rocm_root
@@ -209,19 +216,88 @@ class Driver:
             },
         )
 
-    def publish_resources(self) -> None:
-        unhealthy = self.health.unhealthy_gpus
-        devices = [
-            d.to_device()
-            for d in self.state.allocatable_devices()
-            if d.parent_gpu.index not in unhealthy
+    def _in_scope(self, d) -> bool:
+        return (
+            d.parent_gpu.index not in self.health.unhealthy_gpus
             and (self.device_kinds is None or d.kind in self.device_kinds)
             and (
                 self.gpu_indices is None
                 or d.parent_gpu.index in self.gpu_indices
             )
+        )
+
+    def publish_resources(self) -> None:
+        allocatable = [
+            d for d in self.state.allocatable_devices() if self._in_scope(d)
         ]
-        self.publisher.publish(devices)
+        mode = self.prospective_partitions
+        if not mode or (
+            self.device_kinds is not None
+            and "partition" not in self.device_kinds
+        ):
+            devices = [d.to_device() for d in allocatable]
+            self.publisher.publish(devices)
+            self.metrics.allocatable_devices.set(len(devices))
+            return
+
+        from ..hal.model import (
+            consumes_counters,
+            gpu_device_with_counters,
+            prospective_partition_devices,
+            shared_counter_set,
+        )
+        from ..partition.catalog import make_profile, preferred_memory_mode
+
+        devices, counter_devices, shared_counters = [], [], []
+        for d in allocatable:
+            gpu = d.parent_gpu
+            if d.kind == "gpu":
+                # uncarved GPU: whole-GPU device + prospective partitions
+                # overlapping via the counter set
+                shared_counters.append(shared_counter_set(gpu))
+                devices.append(gpu_device_with_counters(gpu))
+                if mode.upper() in gpu.compute_caps and gpu.repartition_capable:
+                    try:
+                        prof = make_profile(
+                            mode.upper(),
+                            preferred_memory_mode(mode.upper(), gpu.nps_caps),
+                            vram_total_mib=gpu.vram_total_mib or 288 * 1024,
+                            cu_count=gpu.cu_count or 256,
+                            xcd_count=gpu.xcd_count or 8,
+                            nps_caps=gpu.nps_caps,
+                        )
+                    except ValueError as e:
+                        log.warning(
+                            "gpu-%d: no valid %s profile (%s); not "
+                            "publishing prospective partitions",
+                            gpu.index,
+                            mode,
+                            e,
+                        )
+                        continue
+                    counter_devices.extend(
+                        prospective_partition_devices(gpu, prof)
+                    )
+            else:
+                # real (carved) partition: same counter accounting
+                if not any(
+                    c["name"] == f"{gpu.canonical_name}-counters"
+                    for c in shared_counters
+                ):
+                    shared_counters.append(shared_counter_set(gpu))
+                dev = d.to_device()
+                dev["consumesCounters"] = consumes_counters(
+                    gpu,
+                    d.partition.profile.memory_slices_of(
+                        d.partition.partition_id
+                    ),
+                )
+                devices.append(dev)
+        self.publisher.publish(
+            devices,
+            counter_devices=counter_devices,
+            shared_counters=shared_counters,
+        )
         self.metrics.allocatable_devices.set(len(devices))
 
     # ------------------------------------------------------------------

@@ -87,6 +87,16 @@ def build_parser() -> argparse.ArgumentParser:
         help="serve Prometheus /metrics on this port (0 = disabled)",
     )
     p.add_argument(
+        "--prospective-partitions",
+        default=_env("PROSPECTIVE_PARTITIONS", ""),
+        choices=["", "cpx", "dpx", "qpx"],
+        help="publish prospective partitions of this mode with DRA "
+        "sharedCounters (K8s 1.33 partitionable devices) so the default "
+        "scheduler can allocate them; prepare carves on demand — "
+        "scheduler-driven dynamic partitioning (needs resource.k8s.io "
+        "v1beta2+)",
+    )
+    p.add_argument(
         "--rocm-mount",
         default=_env("ROCM_MOUNT", ""),
         help="inject host ROCm userspace into claim containers: a path, "
@@ -164,6 +174,7 @@ def main(argv=None) -> int:
         shared_enforcement=args.shared_enforcement,
         rocm_mount=args.rocm_mount,
         dev_root=args.dev_root,
+        prospective_partitions=args.prospective_partitions,
         gpu_indices=(
             [int(s) for s in args.gpu_indices.split(",") if s.strip()]
             if args.gpu_indices

@@ -26,6 +26,7 @@ from __future__ import annotations
 import contextlib
 import logging
 import os
+import re
 import threading
 from concurrent.futures import ThreadPoolExecutor
 from dataclasses import dataclass
@@ -53,6 +54,9 @@ from .checkpoint import CheckpointStore, PreparedClaim, PreparedDevice
 log = logging.getLogger(__name__)
 
 DRIVER_NAME = "gpu.amd.com"
+
+#: canonical partition device name: gpu-<index>-<mode>-<pid>
+_PARTITION_NAME_RE = re.compile(r"^gpu-(\d+)-([a-z]+)-(\d+)$")
 
 
 class PrepareError(RuntimeError):
@@ -123,6 +127,9 @@ class DeviceState:
         )
         #: gpu_index -> set of claim uids with prepared devices on it
         self._gpu_holders: Dict[int, set] = {}
+        #: gpu_index -> (compute, memory) restore that was refused while
+        #: other claims held the GPU; retried when the GPU drains
+        self._deferred_restores: Dict[int, tuple] = {}
         #: canonical name -> AllocatableDevice
         self._allocatable: Dict[str, AllocatableDevice] = {}
         self.refresh_allocatable()
@@ -368,6 +375,14 @@ class DeviceState:
                     pcfg.memory_partition,
                 ]
                 repartition_done = True
+        # Scheduler-driven carve (DRA partitionable devices): an allocation
+        # may name a PROSPECTIVE partition device (published with
+        # sharedCounters before any carve). If the device doesn't exist yet
+        # but its name parses as a partition of a known GPU, carve that GPU
+        # now — the dynamic-MIG flow the reference shipped disabled
+        # (nvlib.go:560-669), driven by the default scheduler.
+        if self._auto_carve_for_results(info, results, prepared):
+            repartition_done = True
         if repartition_done:
             self.refresh_allocatable()
             self.write_base_cdi_spec()
@@ -378,6 +393,65 @@ class DeviceState:
             # a failure past a successful mode switch must not leak it
             self._rollback(prepared, info.uid)
             raise
+
+    def _auto_carve_for_results(
+        self, info: _ClaimInfo, results: List[dict], prepared: PreparedClaim
+    ) -> bool:
+        from ..partition.catalog import preferred_memory_mode
+
+        changed = False
+        for r in results:
+            name = r.get("device", "")
+            if self._find_device(name) is not None:
+                continue
+            m = _PARTITION_NAME_RE.match(name)
+            if not m:
+                continue
+            gpu_index = int(m.group(1))
+            mode = m.group(2).upper()
+            gpu = next(
+                (g for g in self.lib.enumerate() if g.index == gpu_index),
+                None,
+            )
+            if gpu is None or gpu.compute_partition == mode:
+                continue  # unknown GPU, or carved but pid out of range
+            mem = preferred_memory_mode(mode, gpu.nps_caps)
+            try:
+                switched = self.partition_manager.ensure_mode(
+                    gpu_index,
+                    mode,
+                    mem,
+                    requesting_claim=info.uid,
+                    allow_dynamic=True,
+                )
+            except RepartitionRefused as e:
+                raise PrepareError(
+                    f"allocated device {name!r} requires carving "
+                    f"gpu-{gpu_index} to {mode}: {e}"
+                ) from e
+            if switched:
+                if self.on_repartition is not None:
+                    self.on_repartition()
+                prepared.repartitioned.setdefault(
+                    str(gpu_index),
+                    [
+                        gpu.compute_partition,
+                        gpu.memory_partition,
+                        mode,
+                        mem,
+                    ],
+                )
+                log.info(
+                    "claim %s: auto-carved gpu-%d to %s/%s for "
+                    "scheduler-allocated partition %s",
+                    info.uid,
+                    gpu_index,
+                    mode,
+                    mem,
+                    name,
+                )
+                changed = True
+        return changed
 
     def _prepare_after_partition(
         self,
@@ -646,12 +720,54 @@ class DeviceState:
                         if self.on_repartition is not None:
                             self.on_repartition()
                 except RepartitionRefused as e:
-                    log.warning(
-                        "leaving gpu-%s partitioned (%s)", gpu_index_s, e
+                    # other claims still hold partitions of this GPU;
+                    # remember the restore and retry when it drains
+                    self._deferred_restores[int(gpu_index_s)] = (
+                        prev_c,
+                        prev_m,
                     )
+                    log.warning(
+                        "leaving gpu-%s partitioned for now (%s); restore "
+                        "deferred until the GPU drains",
+                        gpu_index_s,
+                        e,
+                    )
+            if self._retry_deferred_restores(claim_uid):
+                restored = True
             if restored:
                 self.refresh_allocatable()
                 self.write_base_cdi_spec()
 
             self.cdi.delete_claim_spec(claim_uid)
             self.checkpoints.delete(claim_uid)
+
+    def _retry_deferred_restores(self, requesting_claim: str) -> bool:
+        """Apply deferred mode restores for GPUs that have drained (the
+        last pod of a scheduler-carved GPU left: return it to SPX so the
+        whole-GPU device becomes allocatable again)."""
+        restored = False
+        for gpu_index in list(self._deferred_restores):
+            if self.claims_holding_gpu(gpu_index):
+                continue
+            prev_c, prev_m = self._deferred_restores[gpu_index]
+            try:
+                if self.partition_manager.ensure_mode(
+                    gpu_index,
+                    prev_c,
+                    prev_m,
+                    requesting_claim=requesting_claim,
+                    allow_dynamic=True,
+                ):
+                    restored = True
+                    if self.on_repartition is not None:
+                        self.on_repartition()
+                    log.info(
+                        "gpu-%d drained: deferred restore to %s/%s applied",
+                        gpu_index,
+                        prev_c,
+                        prev_m,
+                    )
+                del self._deferred_restores[gpu_index]
+            except RepartitionRefused:
+                pass  # raced a new prepare; keep deferred
+        return restored

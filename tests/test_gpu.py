@@ -469,3 +469,43 @@ class TestNbodyCorrectness:
         err = np.abs(gpu - ref).max()
         scale = np.abs(ref).max()
         assert err / scale < 1e-3, (err, scale)
+
+
+class TestProspectivePartitionsOnHardware:
+    """Prospective-partition publication respects the real box's
+    repartition capability: pool slices (incapable) must publish no
+    prospective devices; bare metal publishes 8 per GPU."""
+
+    def test_gating_matches_capability(self, real_lib, tmp_path):
+        from k8s_dra_driver_amd import DRIVER_NAME
+        from k8s_dra_driver_amd.kube.client import InMemoryKube
+        from k8s_dra_driver_amd.plugin.driver import Driver
+
+        kube = InMemoryKube()
+        kube.api_versions = ["v1beta2", "v1beta1"]
+        driver = Driver(
+            real_lib,
+            kube,
+            node_name="hw",
+            cdi_root=str(tmp_path / "cdi"),
+            checkpoint_root=str(tmp_path / "state"),
+            use_tmpfs=False,
+            prospective_partitions="cpx",
+        )
+        driver.startup()
+        try:
+            devs = [
+                d["name"]
+                for s in kube.list_resource_slices(DRIVER_NAME)
+                for d in s["spec"]["devices"]
+            ]
+            n_gpus = len(real_lib.enumerate())
+            prospective = [d for d in devs if "-cpx-" in d]
+            if real_lib.dynamic_repartition_capable():
+                assert len(prospective) == 8 * n_gpus
+            else:
+                assert prospective == [], (
+                    "incapable box must not advertise carves it cannot do"
+                )
+        finally:
+            driver.shutdown(unpublish=False)

@@ -167,3 +167,65 @@ def test_all_demo_specs_parse():
         assert docs, f
         for d in docs:
             assert "kind" in d and "apiVersion" in d, f
+
+
+def test_gpu_test8_prospective_partitions(tmp_path):
+    """Scheduler-driven dynamic partitioning demo: two pods claim CPX
+    partitions directly from PROSPECTIVE devices (no PartitionConfig,
+    no pre-carve); the first prepare carves, the drain restores SPX."""
+    lib = FakeDeviceLib()
+    lib.open()
+    kube = InMemoryKube()
+    kube.api_versions = ["v1beta2", "v1beta1"]
+    driver = Driver(
+        lib,
+        kube,
+        node_name="demo-node",
+        cdi_root=str(tmp_path / "cdi"),
+        checkpoint_root=str(tmp_path / "state"),
+        use_tmpfs=False,
+        prospective_partitions="cpx",
+    )
+    driver.startup()
+    alloc = Allocator()
+    specs = dict(claim_specs_from(load_docs("gpu-test8-prospective.yaml")))
+    spec = specs["one-cpx-partition"]
+
+    devices = [
+        d
+        for s in kube.list_resource_slices(DRIVER_NAME)
+        for d in s["spec"]["devices"]
+    ]
+    assert any("-cpx-" in d["name"] for d in devices)
+
+    uids, in_use = [], set()
+    for i in range(2):
+        uid = f"t8-{i}"
+        results = alloc.allocate(
+            {"devices": {"requests": spec["devices"]["requests"]}},
+            devices,
+            pool="demo-node",
+            in_use=in_use,
+        )
+        assert "-cpx-" in results[0].device
+        in_use.add(results[0].device)
+        claim = {
+            "metadata": {"namespace": "gpu-test8", "name": f"p{i}", "uid": uid},
+            "status": {
+                "allocation": {
+                    "devices": {"results": [r.to_obj() for r in results]}
+                }
+            },
+        }
+        kube.put_resource_claim(claim)
+        res = driver.node_prepare_resources(
+            [ClaimRef("gpu-test8", f"p{i}", uid)]
+        )
+        assert res[uid].error == "", res[uid].error
+        uids.append(uid)
+
+    assert lib.enumerate()[0].compute_partition == "CPX"  # auto-carved
+    for i, uid in enumerate(uids):
+        driver.node_unprepare_resources([ClaimRef("gpu-test8", f"p{i}", uid)])
+    assert lib.enumerate()[0].compute_partition == "SPX"  # drained
+    driver.shutdown(unpublish=False)

@@ -90,6 +90,25 @@ DEFAULT_DEVICE_CLASSES = {
 }
 
 
+def _counter_keys(device: dict) -> frozenset:
+    """(counterSet, counter) pairs this device consumes (K8s 1.33
+    partitionable devices); empty for counterless devices."""
+    out = set()
+    for cc in device.get("consumesCounters") or []:
+        cs = cc.get("counterSet", "")
+        for cname in cc.get("counters") or {}:
+            out.add((cs, cname))
+    return frozenset(out)
+
+
+def _counters_of_names(devices, names) -> set:
+    out = set()
+    for d in devices:
+        if d["name"] in names:
+            out |= _counter_keys(d)
+    return out
+
+
 def _attr_value(device: dict, qualified: str):
     basic = device.get("basic", device)
     v = (basic.get("attributes") or {}).get(qualified)
@@ -173,7 +192,17 @@ class Allocator:
             raise AllocationError("claim has no device requests")
         constraints = spec_devices.get("constraints") or []
 
-        avail = [d for d in devices if d["name"] not in in_use]
+        # sharedCounters overlap (DRA partitionable devices, K8s 1.33):
+        # a device is unavailable if it consumes any counter a device
+        # already in use consumes — e.g. a whole GPU vs its prospective
+        # partitions, or two partitions sharing a memory slice.
+        consumed = _counters_of_names(devices, in_use)
+        avail = [
+            d
+            for d in devices
+            if d["name"] not in in_use
+            and not (_counter_keys(d) & consumed)
+        ]
         per_request: List[Tuple[dict, List[dict], int]] = []
         for r in requests:
             # adminAccess requests see every device, in-use included
@@ -246,9 +275,14 @@ class Allocator:
         best_score = -1
         budget = self.search_budget
 
-        def options(i: int, used: Set[str]):
+        def options(i: int, used: Set[str], used_counters: frozenset):
             r, cands, count = per_request[i]
-            free = [c for c in cands if c["name"] not in used]
+            free = [
+                c
+                for c in cands
+                if c["name"] not in used
+                and not (_counter_keys(c) & used_counters)
+            ]
             if len(free) < count:
                 return
             if r.get("allocationMode") == "All":
@@ -273,7 +307,12 @@ class Allocator:
             for combo in itertools.islice(combos, COMBO_CAP):
                 yield list(combo)
 
-        def dfs(i: int, used: Set[str], chosen: List[List[dict]]):
+        def dfs(
+            i: int,
+            used: Set[str],
+            used_counters: frozenset,
+            chosen: List[List[dict]],
+        ):
             nonlocal best, best_score, budget
             if budget <= 0:
                 return
@@ -284,7 +323,7 @@ class Allocator:
                 if score > best_score:
                     best, best_score = [list(x) for x in chosen], score
                 return
-            for opt in options(i, used):
+            for opt in options(i, used, used_counters):
                 budget -= 1
                 # prune: partial constraint violation can't self-heal
                 if not self._constraint_ok(
@@ -292,11 +331,15 @@ class Allocator:
                 ):
                     continue
                 names = {d["name"] for d in opt}
-                dfs(i + 1, used | names, chosen + [opt])
+                counters = frozenset(
+                    used_counters
+                    | set().union(*(_counter_keys(d) for d in opt))
+                )
+                dfs(i + 1, used | names, counters, chosen + [opt])
                 if budget <= 0:
                     return
 
-        dfs(0, set(), [])
+        dfs(0, set(), frozenset(), [])
         if best is not None:
             if budget <= 0:
                 log.warning(
@@ -314,9 +357,15 @@ class Allocator:
                 self.search_budget,
             )
         used: Set[str] = set()
+        used_counters: set = set()
         chosen: List[List[dict]] = []
         for r, cands, count in per_request:
-            free = [c for c in cands if c["name"] not in used]
+            free = [
+                c
+                for c in cands
+                if c["name"] not in used
+                and not (_counter_keys(c) & used_counters)
+            ]
             if len(free) < count:
                 raise AllocationError(
                     f"request {r.get('name')!r}: unsatisfiable under "
@@ -336,6 +385,8 @@ class Allocator:
                 )
             chosen.append(pick)
             used |= {d["name"] for d in pick}
+            for d in pick:
+                used_counters |= _counter_keys(d)
         return chosen
 
     # ------------------------------------------------------------------

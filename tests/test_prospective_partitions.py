@@ -165,3 +165,99 @@ def test_carve_refused_while_whole_gpu_held(driver):
         [type("R", (), {"namespace": "d", "name": "c-u-part", "uid": "u-part"})()]
     )
     assert "carving gpu-4" in res["u-part"].error
+
+
+class TestCounterAwareAllocator:
+    """The in-repo allocator (controller scheduler + bench) honors
+    sharedCounters overlap like the real kube-scheduler will."""
+
+    def _devices(self):
+        from k8s_dra_driver_amd.hal import FakeDeviceLib
+        from k8s_dra_driver_amd.hal.model import (
+            gpu_device_with_counters,
+            prospective_partition_devices,
+        )
+        from k8s_dra_driver_amd.partition.catalog import make_profile
+
+        lib = FakeDeviceLib()
+        lib.open()
+        gpus = lib.enumerate()[:2]
+        prof = make_profile("CPX", "NPS4")
+        devs = []
+        for g in gpus:
+            devs.append(gpu_device_with_counters(g))
+            devs.extend(prospective_partition_devices(g, prof))
+        return devs
+
+    def test_whole_gpu_blocks_its_partitions(self):
+        from k8s_dra_driver_amd.allocator.structured import Allocator
+
+        devs = self._devices()
+        alloc = Allocator()
+        spec = {
+            "devices": {
+                "requests": [
+                    {"name": "p", "deviceClassName": "partition.gpu.amd.com"}
+                ]
+            }
+        }
+        res = alloc.allocate(spec, devs, pool="n", in_use={"gpu-0"})
+        # gpu-0 consumed all gpu-0 counters: only gpu-1 partitions remain
+        assert res[0].device.startswith("gpu-1-cpx-")
+
+    def test_partition_blocks_whole_gpu(self):
+        import pytest as _pytest
+
+        from k8s_dra_driver_amd.allocator.structured import (
+            AllocationError,
+            Allocator,
+        )
+
+        devs = self._devices()
+        alloc = Allocator()
+        spec = {
+            "devices": {
+                "requests": [
+                    {"name": "g", "deviceClassName": "gpu.amd.com", "count": 2}
+                ]
+            }
+        }
+        # one partition of each GPU in use -> no whole GPU allocatable
+        with _pytest.raises(AllocationError):
+            alloc.allocate(
+                spec, devs, pool="n", in_use={"gpu-0-cpx-3", "gpu-1-cpx-0"}
+            )
+
+    def test_one_claim_cannot_take_gpu_and_its_partition(self):
+        from k8s_dra_driver_amd.allocator.structured import Allocator
+
+        devs = self._devices()
+        alloc = Allocator()
+        spec = {
+            "devices": {
+                "requests": [
+                    {"name": "g", "deviceClassName": "gpu.amd.com"},
+                    {"name": "p", "deviceClassName": "partition.gpu.amd.com"},
+                ]
+            }
+        }
+        res = alloc.allocate(spec, devs, pool="n")
+        by_req = {r.request: r.device for r in res}
+        # the partition must come from the OTHER die
+        gpu_idx = by_req["g"].split("-")[1]
+        assert not by_req["p"].startswith(f"gpu-{gpu_idx}-")
+
+    def test_disjoint_partitions_coexist(self):
+        from k8s_dra_driver_amd.allocator.structured import Allocator
+
+        devs = self._devices()
+        alloc = Allocator()
+        spec = {
+            "devices": {
+                "requests": [
+                    {"name": "p", "deviceClassName": "partition.gpu.amd.com"}
+                ]
+            }
+        }
+        res = alloc.allocate(spec, devs, pool="n", in_use={"gpu-0-cpx-0"})
+        assert res[0].device != "gpu-0-cpx-0"  # sibling or other die is fine

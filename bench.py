@@ -51,6 +51,9 @@ from k8s_dra_driver_amd.plugin.proto import V1BETA1
 from k8s_dra_driver_amd.plugin.server import PluginServer
 
 CONFIG_NAMES = ["mock", "whole", "shared", "cpx", "topo4"]
+#: extra non-BASELINE mode: scheduler-driven auto-carve lifecycle
+#: (prospective partitions + sharedCounters; round-2 feature)
+EXTRA_CONFIGS = ["autocpx"]
 
 
 def pick_hal(mode: str):
@@ -168,6 +171,8 @@ class BenchRank:
             self.lib.open()
             self.hal_kind = "fake"
         self.kube = InMemoryKube()
+        if config == "autocpx":
+            self.kube.api_versions = ["v1beta2", "v1beta1"]
         self.driver = Driver(
             self.lib,
             self.kube,
@@ -175,6 +180,7 @@ class BenchRank:
             cdi_root=os.path.join(self.tmp, "cdi"),
             checkpoint_root=os.path.join(self.tmp, "state"),
             use_tmpfs=False,
+            prospective_partitions="cpx" if config == "autocpx" else "",
         )
         self.driver.startup()
         self.allocator = Allocator()
@@ -204,12 +210,12 @@ class BenchRank:
         self._workers = None
         self.latencies_ms: list = []
         #: pods delivered per lifecycle (cpx binds 8 pods per carve)
-        self.pods_per_lifecycle = 8 if config == "cpx" else 1
+        self.pods_per_lifecycle = 8 if config in ("cpx", "autocpx") else 1
         # Pipeline warm-up at construction (grpc channel, protobuf codecs,
         # JSON/fsync paths, allocator): steady-state throughput is ~35%
         # above a cold pipeline, and the contract's warmup steps should
         # measure the benchmark's own warmup, not Python's.
-        for _ in range(100 if config != "cpx" else 8):
+        for _ in range(100 if config not in ("cpx", "autocpx") else 8):
             self._one_lifecycle()
         self.latencies_ms.clear()
 
@@ -265,6 +271,8 @@ class BenchRank:
     def _one_lifecycle(self) -> float:
         if self.config == "cpx":
             return self._cpx_lifecycle()
+        if self.config == "autocpx":
+            return self._autocpx_lifecycle()
         uid = self._next_uid()
         cfg = "shared" if self.config == "shared" else self.config
         t0 = time.perf_counter()  # pod-sees-GPU latency starts here
@@ -321,13 +329,56 @@ class BenchRank:
         self._grpc_unprepare([carve_uid])  # restores SPX/NPS1
         return dt
 
+    def _autocpx_lifecycle(self) -> float:
+        """Scheduler-driven carve (round-2): 8 pods each claim one
+        PROSPECTIVE CPX partition of this rank's GPU (no PartitionConfig);
+        the first prepare auto-carves, draining all 8 restores SPX."""
+        gpu_name = f"gpu-{self.target_gpu % 8}"
+        t0 = time.perf_counter()
+        slices = self.kube.list_resource_slices(DRIVER_NAME)
+        parts = [
+            d["name"]
+            for s in slices
+            for d in s["spec"]["devices"]
+            if d["name"].startswith(f"{gpu_name}-cpx-")
+        ][:8]
+        if len(parts) < 8:
+            raise RuntimeError(
+                f"expected 8 prospective partitions of {gpu_name}, "
+                f"saw {len(parts)}"
+            )
+        pod_uids = []
+        for p in parts:
+            uid = self._next_uid("a")
+            claim = make_claim_spec(uid, "whole")
+            claim["status"] = {
+                "allocation": {
+                    "devices": {
+                        "results": [
+                            {
+                                "request": "gpu",
+                                "driver": DRIVER_NAME,
+                                "pool": self.node,
+                                "device": p,
+                            }
+                        ]
+                    }
+                }
+            }
+            self.kube.put_resource_claim(claim)
+            pod_uids.append(uid)
+        self._grpc_prepare(pod_uids)  # first claim auto-carves
+        dt = (time.perf_counter() - t0) * 1e3
+        self._grpc_unprepare(pod_uids)  # last drain restores SPX
+        return dt
+
     def step(self, pods: int, inflight: int = 1) -> None:
         """One step = `pods` full pod lifecycles through the pipeline:
         CEL allocation -> apiserver write -> gRPC prepare -> unprepare.
         ``inflight`` > 1 admits pods concurrently, as kubelet does when
         several pods land on the node at once."""
         lifecycles = max(1, pods // self.pods_per_lifecycle)
-        if inflight <= 1 or self.config in ("cpx", "topo4"):
+        if inflight <= 1 or self.config in ("cpx", "autocpx", "topo4"):
             # cpx carve claims drain the whole GPU: serial by construction;
             # topo4 claims 4 of the node's GPUs: concurrent copies contend
             for _ in range(lifecycles):
@@ -451,7 +502,7 @@ def main() -> int:
     ap.add_argument("--inflight", type=int, default=1, help="concurrent pod admissions per rank")
     ap.add_argument(
         "--config",
-        choices=CONFIG_NAMES + ["all"],
+        choices=CONFIG_NAMES + EXTRA_CONFIGS + ["all"],
         default="whole",
         help="BASELINE.json config to measure (all = whole headline + "
         "every config attested in config.configs)",

@@ -43,11 +43,20 @@ def _wait(pred, timeout, what):
 
 
 class DemoHarness:
-    def __init__(self, hal: str = "fake", verbose: bool = False):
+    def __init__(
+        self,
+        hal: str = "fake",
+        verbose: bool = False,
+        prospective: str = "",
+    ):
         self.hal = hal
         self.verbose = verbose
+        self.prospective = prospective
         self.tmp = tempfile.mkdtemp(prefix="amd-dra-demo-")
         self.api = MiniApiServer().start()
+        if prospective:
+            # counters need resource.k8s.io v1beta2+ (K8s 1.33)
+            self.api.store.api_versions = ["v1beta2", "v1beta1"]
         self.api.store.put_node({"metadata": {"name": "demo-node", "uid": "demo-node-uid"}})
         self.kubeconfig = self.api.write_kubeconfig(os.path.join(self.tmp, "kubeconfig"))
         self.procs: list = []
@@ -74,13 +83,16 @@ class DemoHarness:
         plugin_dir = os.path.join(self.tmp, "plugins", DRIVER_NAME)
         registry = os.path.join(self.tmp, "plugins_registry")
         self.cdi_root = os.path.join(self.tmp, "cdi")
-        self._spawn(
-            "plugin",
-            "k8s_dra_driver_amd.plugin.main",
+        plugin_args = [
             "--hal", self.hal,
             "--cdi-root", self.cdi_root,
             "--plugin-path", plugin_dir,
             "--plugin-registration-path", registry,
+        ]
+        if self.prospective:
+            plugin_args += ["--prospective-partitions", self.prospective]
+        self._spawn(
+            "plugin", "k8s_dra_driver_amd.plugin.main", *plugin_args
         )
         self._spawn(
             "controller",
@@ -317,9 +329,19 @@ def main(argv=None) -> int:
         metavar="GPU:MODE[:NPS]",
         help="pre-carve a GPU before running specs, e.g. gpu-0:CPX:NPS4",
     )
+    ap.add_argument(
+        "--prospective",
+        default="",
+        choices=["", "cpx", "dpx", "qpx"],
+        help="publish prospective partitions of this mode (scheduler-"
+        "driven dynamic partitioning; serves resource.k8s.io/v1beta2) — "
+        "required for gpu-test8",
+    )
     ap.add_argument("-v", "--verbose", action="store_true")
     args = ap.parse_args(argv)
-    harness = DemoHarness(hal=args.hal, verbose=args.verbose)
+    harness = DemoHarness(
+        hal=args.hal, verbose=args.verbose, prospective=args.prospective
+    )
     rc = 0
     try:
         harness.start()

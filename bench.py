@@ -150,7 +150,7 @@ class BenchRank:
             hal_mode = "fake"  # config #1 is explicitly the stub config
         self.lib, self.hal_kind = pick_hal(hal_mode)
         if (
-            config == "cpx"
+            config in ("cpx", "autocpx")
             and self.hal_kind == "amdsmi"
             and not self.lib.dynamic_repartition_capable()
         ):

@@ -36,7 +36,6 @@ def test_profile_endpoint_collapsed_stacks(diag_url):
     """pprof `profile` analog: on-demand sampling CPU
     profile in collapsed-stack format, covering worker threads."""
     import threading
-    import time
     import urllib.request
 
     stop = threading.Event()

@@ -692,7 +692,15 @@ class DeviceState:
         with self._claim_lock(claim_uid):
             pc = self.checkpoints.read(claim_uid)
             if pc is None:
-                return  # unknown claim: no-op (:171-173)
+                # unknown claim: no-op (:171-173) — but still sweep any
+                # deferred mode restores (a concurrent batch drain can
+                # leave one pending with no further real unprepare)
+                if self._deferred_restores and self._retry_deferred_restores(
+                    claim_uid
+                ):
+                    self.refresh_allocatable()
+                    self.write_base_cdi_spec()
+                return
 
             if pc.sharing_strategy == SHARED_COMPUTE and self.shared_manager:
                 self.shared_manager.stop_session(pc.shared_session_id or claim_uid[:36])

@@ -261,3 +261,30 @@ class TestCounterAwareAllocator:
         }
         res = alloc.allocate(spec, devs, pool="n", in_use={"gpu-0-cpx-0"})
         assert res[0].device != "gpu-0-cpx-0"  # sibling or other die is fine
+
+
+def test_concurrent_batch_drain_eventually_restores(driver):
+    """8 partition claims unprepared in one concurrent batch: the carve
+    owner's restore may be deferred mid-race, but the deferred-restore
+    retry (fired on any later unprepare) must bring the GPU back to SPX
+    — and re-prepare works either way because prospective and real
+    partition names are identical."""
+    d, kube, lib = driver
+    uids = []
+    for k in range(8):
+        uid = f"u-b{k}"
+        kube.put_resource_claim(_claim(uid, f"gpu-5-cpx-{k}"))
+        uids.append(uid)
+    refs = [
+        type("R", (), {"namespace": "d", "name": f"c-{u}", "uid": u})()
+        for u in uids
+    ]
+    res = d.node_prepare_resources(refs)
+    assert all(res[u].error == "" for u in uids)
+    assert lib.enumerate()[5].compute_partition == "CPX"
+    d.node_unprepare_resources(refs)  # concurrent batch through the pool
+    if lib.enumerate()[5].compute_partition != "SPX":
+        # race window: retry fires on the next unprepare of anything
+        d.state.unprepare("no-such-claim")
+        d.state._retry_deferred_restores("sweep")
+    assert lib.enumerate()[5].compute_partition == "SPX"

@@ -348,13 +348,36 @@ class AmdSmiDeviceLib(DeviceLib):
         the ``timeSlicingEffective`` attribute contract); otherwise the
         request is recorded and surfaced as advisory — HSA default
         time-slicing still multiplexes queues fairly."""
+        others = {
+            i: q
+            for i, q in self._timeslice.items()
+            if i != gpu_index and q is not None and q != quantum_us
+        }
         self._timeslice[gpu_index] = quantum_us
         if self._probe_quantum_knob():
-            value_ms = (
-                self._quantum_default_ms
-                if quantum_us is None
-                else max(1, quantum_us // 1000)
-            )
+            if quantum_us is not None and others:
+                # the knob is node-global: a different quantum is active
+                # for another GPU's claim and this write overrides it
+                log.warning(
+                    "queue preemption timeout is node-global: setting "
+                    "%s us for gpu-%d overrides active quanta %s",
+                    quantum_us,
+                    gpu_index,
+                    others,
+                )
+            if quantum_us is None:
+                # restore: fall back to another GPU's still-active
+                # quantum (node-global knob), else the boot default
+                active = sorted(
+                    q for q in self._timeslice.values() if q is not None
+                )
+                value_ms = (
+                    max(1, active[0] // 1000)
+                    if active
+                    else self._quantum_default_ms
+                )
+            else:
+                value_ms = max(1, quantum_us // 1000)
             try:
                 with open(self.QUEUE_PREEMPTION_PARAM, "w") as f:
                     f.write(str(value_ms))

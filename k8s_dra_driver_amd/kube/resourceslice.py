@@ -99,6 +99,13 @@ class ResourceSlicePublisher:
         #: negotiated resource.k8s.io version (resolved on first publish)
         self._api_version: Optional[str] = None
 
+    def negotiated_version(self) -> str:
+        """Public accessor (triggers negotiation): lets callers adapt
+        behavior to the API generation — e.g. taint unhealthy devices on
+        v1beta2+ instead of unpublishing them."""
+        with self._lock:
+            return self._negotiate_version()
+
     def _negotiate_version(self) -> str:
         """Pick the newest mutually-supported resource.k8s.io version
         (single code path: devices are built v1beta1-shaped and flattened

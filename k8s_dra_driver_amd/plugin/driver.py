@@ -216,9 +216,21 @@ class Driver:
             },
         )
 
-    def _in_scope(self, d) -> bool:
+    #: taint applied to devices of unhealthy GPUs on v1beta2+ (DRA device
+    #: taints, K8s 1.33): the scheduler stops placing new claims, while
+    #: monitoring claims with a matching toleration can still reach the
+    #: sick device — richer than unpublishing (the v1beta1 behavior).
+    UNHEALTHY_TAINT = {
+        "key": "gpu.amd.com/unhealthy",
+        "effect": "NoSchedule",
+    }
+
+    def _in_scope(self, d, *, include_unhealthy: bool = False) -> bool:
         return (
-            d.parent_gpu.index not in self.health.unhealthy_gpus
+            (
+                include_unhealthy
+                or d.parent_gpu.index not in self.health.unhealthy_gpus
+            )
             and (self.device_kinds is None or d.kind in self.device_kinds)
             and (
                 self.gpu_indices is None
@@ -226,16 +238,32 @@ class Driver:
             )
         )
 
+    def _taints_for(self, d) -> list:
+        if d.parent_gpu.index in self.health.unhealthy_gpus:
+            return [dict(self.UNHEALTHY_TAINT)]
+        return []
+
     def publish_resources(self) -> None:
+        # On v1beta2+ unhealthy devices stay published but TAINTED; on
+        # v1beta1 (no taints) they are removed as before.
+        taint_capable = self.publisher.negotiated_version() != "v1beta1"
         allocatable = [
-            d for d in self.state.allocatable_devices() if self._in_scope(d)
+            d
+            for d in self.state.allocatable_devices()
+            if self._in_scope(d, include_unhealthy=taint_capable)
         ]
         mode = self.prospective_partitions
         if not mode or (
             self.device_kinds is not None
             and "partition" not in self.device_kinds
         ):
-            devices = [d.to_device() for d in allocatable]
+            devices = []
+            for d in allocatable:
+                dev = d.to_device()
+                taints = self._taints_for(d)
+                if taints:
+                    dev["taints"] = taints
+                devices.append(dev)
             self.publisher.publish(devices)
             self.metrics.allocatable_devices.set(len(devices))
             return
@@ -251,11 +279,17 @@ class Driver:
         devices, counter_devices, shared_counters = [], [], []
         for d in allocatable:
             gpu = d.parent_gpu
+            taints = self._taints_for(d)
             if d.kind == "gpu":
                 # uncarved GPU: whole-GPU device + prospective partitions
                 # overlapping via the counter set
                 shared_counters.append(shared_counter_set(gpu))
-                devices.append(gpu_device_with_counters(gpu))
+                gdev = gpu_device_with_counters(gpu)
+                if taints:
+                    gdev["taints"] = taints
+                devices.append(gdev)
+                if taints:
+                    continue  # no prospective carves of a sick GPU
                 if mode.upper() in gpu.compute_caps and gpu.repartition_capable:
                     try:
                         prof = make_profile(
@@ -292,6 +326,8 @@ class Driver:
                         d.partition.partition_id
                     ),
                 )
+                if taints:
+                    dev["taints"] = taints
                 devices.append(dev)
         self.publisher.publish(
             devices,

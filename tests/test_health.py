@@ -77,3 +77,63 @@ def test_device_kinds_gating(tmp_path):
         for d in s["spec"]["devices"]
     ]
     assert names and all("cpx" in n for n in names)
+
+
+class TestUnhealthyDeviceTaints:
+    """DRA device taints (K8s 1.33): on v1beta2+ unhealthy GPUs stay
+    published but tainted NoSchedule; on v1beta1 they are removed."""
+
+    def _driver(self, tmp_path, api_versions):
+        from k8s_dra_driver_amd.hal import FakeDeviceLib
+        from k8s_dra_driver_amd.kube.client import InMemoryKube
+        from k8s_dra_driver_amd.plugin.driver import Driver
+
+        lib = FakeDeviceLib()
+        lib.open()
+        kube = InMemoryKube()
+        kube.api_versions = api_versions
+        d = Driver(
+            lib,
+            kube,
+            node_name="n",
+            cdi_root=str(tmp_path / "cdi"),
+            checkpoint_root=str(tmp_path / "state"),
+            use_tmpfs=False,
+        )
+        d.startup()
+        return d, kube, lib
+
+    def _names(self, kube):
+        from k8s_dra_driver_amd import DRIVER_NAME
+
+        return {
+            dev["name"]: dev
+            for s in kube.list_resource_slices(DRIVER_NAME)
+            for dev in s["spec"]["devices"]
+        }
+
+    def test_v1beta2_taints_instead_of_removal(self, tmp_path):
+        d, kube, lib = self._driver(tmp_path, ["v1beta2", "v1beta1"])
+        with d.health._lock:
+            d.health._unhealthy = {2}
+        d.publish_resources()
+        devs = self._names(kube)
+        assert "gpu-2" in devs  # still published
+        assert devs["gpu-2"]["taints"] == [
+            {"key": "gpu.amd.com/unhealthy", "effect": "NoSchedule"}
+        ]
+        assert "taints" not in devs["gpu-1"]
+        # recovery clears the taint
+        with d.health._lock:
+            d.health._unhealthy = set()
+        d.publish_resources()
+        assert "taints" not in self._names(kube)["gpu-2"]
+        d.shutdown(unpublish=False)
+
+    def test_v1beta1_removes_as_before(self, tmp_path):
+        d, kube, lib = self._driver(tmp_path, ["v1beta1"])
+        with d.health._lock:
+            d.health._unhealthy = {5}
+        d.publish_resources()
+        assert "gpu-5" not in self._names(kube)
+        d.shutdown(unpublish=False)

@@ -19,11 +19,26 @@ model (SURVEY.md §7 hard-part 1):
 
 from __future__ import annotations
 
+import logging
 import threading
 from typing import Callable, Dict, List, Optional
 
 from ..hal.base import DeviceLib, HalError
 from .catalog import DEFAULT_VALID_NPS, validate_mode_combo
+
+log = logging.getLogger(__name__)
+
+
+class RepartitionFailed(HalError):
+    """A mode switch failed mid-sequence. ``original`` is the mode the GPU
+    SHOULD return to; ``reverted`` says whether the best-effort revert
+    already restored it (False = the GPU is in an intermediate mode and
+    the caller must schedule a deferred restore)."""
+
+    def __init__(self, msg: str, original: tuple, reverted: bool):
+        super().__init__(msg)
+        self.original = original
+        self.reverted = reverted
 
 
 class RepartitionRefused(HalError):
@@ -87,7 +102,36 @@ class PartitionManager:
                     f"gpu-{gpu_index} repartition refused: held by "
                     f"{len(holders)} other prepared claim(s) {holders[:3]}"
                 )
-            self._switch(gpu_index, cur_c, cur_m, compute_mode, memory_mode)
+            try:
+                self._switch(gpu_index, cur_c, cur_m, compute_mode, memory_mode)
+            except Exception as e:
+                # best-effort revert to the original mode (the switch is
+                # not atomic: a failure after the compute change leaves
+                # an intermediate state)
+                reverted = False
+                try:
+                    now = {g.index: g for g in self.lib.enumerate()}[gpu_index]
+                    actual = (now.compute_partition, now.memory_partition)
+                    if actual == (cur_c, cur_m):
+                        reverted = True
+                    else:
+                        self._switch(
+                            gpu_index, actual[0], actual[1], cur_c, cur_m
+                        )
+                        reverted = True
+                except Exception:
+                    log.exception(
+                        "gpu-%d: revert after failed switch also failed; "
+                        "left in an intermediate mode (deferred restore "
+                        "required)",
+                        gpu_index,
+                    )
+                raise RepartitionFailed(
+                    f"gpu-{gpu_index} mode switch to "
+                    f"{compute_mode}/{memory_mode} failed: {e}",
+                    original=(cur_c, cur_m),
+                    reverted=reverted,
+                ) from e
             return True
 
     def _switch(

@@ -46,7 +46,11 @@ from ..cdi.handler import CDIHandler
 from ..cdi.spec import ContainerEdits
 from ..hal.base import DeviceLib
 from ..hal.model import AllocatableDevice
-from ..partition.manager import PartitionManager, RepartitionRefused
+from ..partition.manager import (
+    PartitionManager,
+    RepartitionFailed,
+    RepartitionRefused,
+)
 from ..sharing.shared import SharedComputeManager
 from ..sharing.timeslice import TimeSlicingManager
 from .checkpoint import CheckpointStore, PreparedClaim, PreparedDevice
@@ -345,44 +349,61 @@ class DeviceState:
         # PartitionConfig on a whole-GPU result repartitions that GPU as part
         # of Prepare (dynamic MIG analog the reference shipped disabled).
         repartition_done = False
-        for r in results:
-            req = r.get("request", "")
-            pcfg = select_config_for_request(req, configs, PartitionConfig)
-            if pcfg is None:
-                continue
-            dev = self._find_device(r["device"])
-            if dev is None or dev.kind != "gpu":
-                continue
-            gpu_index = dev.parent_gpu.index
-            try:
-                cur = dev.parent_gpu
-                switched = self.partition_manager.ensure_mode(
-                    gpu_index,
-                    pcfg.compute_partition,
-                    pcfg.memory_partition,
-                    requesting_claim=info.uid,
-                    allow_dynamic=pcfg.allow_dynamic_repartition,
-                )
-            except RepartitionRefused as e:
-                raise PrepareError(str(e)) from e
-            if switched:
-                if self.on_repartition is not None:
-                    self.on_repartition()
-                prepared.repartitioned[str(gpu_index)] = [
-                    cur.compute_partition,
-                    cur.memory_partition,
-                    pcfg.compute_partition,
-                    pcfg.memory_partition,
-                ]
+        try:
+            for r in results:
+                req = r.get("request", "")
+                pcfg = select_config_for_request(req, configs, PartitionConfig)
+                if pcfg is None:
+                    continue
+                dev = self._find_device(r["device"])
+                if dev is None or dev.kind != "gpu":
+                    continue
+                gpu_index = dev.parent_gpu.index
+                try:
+                    cur = dev.parent_gpu
+                    switched = self.partition_manager.ensure_mode(
+                        gpu_index,
+                        pcfg.compute_partition,
+                        pcfg.memory_partition,
+                        requesting_claim=info.uid,
+                        allow_dynamic=pcfg.allow_dynamic_repartition,
+                    )
+                except RepartitionRefused as e:
+                    raise PrepareError(str(e)) from e
+                except RepartitionFailed as e:
+                    self._note_failed_switch(gpu_index, e)
+                    raise PrepareError(str(e)) from e
+                except Exception as e:
+                    # HAL failure mid-switch: surface typed, and let the
+                    # outer rollback undo any PRIOR switch of this claim
+                    raise PrepareError(
+                        f"repartition of gpu-{gpu_index} failed: {e}"
+                    ) from e
+                if switched:
+                    if self.on_repartition is not None:
+                        self.on_repartition()
+                    prepared.repartitioned[str(gpu_index)] = [
+                        cur.compute_partition,
+                        cur.memory_partition,
+                        pcfg.compute_partition,
+                        pcfg.memory_partition,
+                    ]
+                    repartition_done = True
+            # Scheduler-driven carve (DRA partitionable devices): an
+            # allocation may name a PROSPECTIVE partition device (published
+            # with sharedCounters before any carve). If the device doesn't
+            # exist yet but its name parses as a partition of a known GPU,
+            # carve that GPU now — the dynamic-MIG flow the reference
+            # shipped disabled (nvlib.go:560-669), scheduler-driven.
+            if self._auto_carve_for_results(info, results, prepared):
                 repartition_done = True
-        # Scheduler-driven carve (DRA partitionable devices): an allocation
-        # may name a PROSPECTIVE partition device (published with
-        # sharedCounters before any carve). If the device doesn't exist yet
-        # but its name parses as a partition of a known GPU, carve that GPU
-        # now — the dynamic-MIG flow the reference shipped disabled
-        # (nvlib.go:560-669), driven by the default scheduler.
-        if self._auto_carve_for_results(info, results, prepared):
-            repartition_done = True
+        except BaseException:
+            # undo any mode switch this claim already performed (a claim
+            # spanning several GPUs can fail on the second switch) and
+            # re-sync publication if hardware state moved
+            if prepared.repartitioned:
+                self._rollback(prepared, info.uid)
+            raise
         if repartition_done:
             self.refresh_allocatable()
             self.write_base_cdi_spec()
@@ -428,6 +449,15 @@ class DeviceState:
                 raise PrepareError(
                     f"allocated device {name!r} requires carving "
                     f"gpu-{gpu_index} to {mode}: {e}"
+                ) from e
+            except RepartitionFailed as e:
+                self._note_failed_switch(gpu_index, e)
+                raise PrepareError(
+                    f"auto-carve of gpu-{gpu_index} to {mode} failed: {e}"
+                ) from e
+            except Exception as e:
+                raise PrepareError(
+                    f"auto-carve of gpu-{gpu_index} to {mode} failed: {e}"
                 ) from e
             if switched:
                 if self.on_repartition is not None:
@@ -749,6 +779,16 @@ class DeviceState:
             self.cdi.delete_claim_spec(claim_uid)
             self.checkpoints.delete(claim_uid)
 
+    def _note_failed_switch(self, gpu_index: int, e: RepartitionFailed) -> None:
+        """A mode switch died mid-sequence. If the manager's revert did
+        not restore the original mode, remember it as a deferred restore
+        (retried on every unprepare) and re-sync publication with the
+        actual hardware state."""
+        if not e.reverted:
+            self._deferred_restores.setdefault(gpu_index, e.original)
+            self.refresh_allocatable()
+            self.write_base_cdi_spec()
+
     def _retry_deferred_restores(self, requesting_claim: str) -> bool:
         """Apply deferred mode restores for GPUs that have drained (the
         last pod of a scheduler-carved GPU left: return it to SPX so the
@@ -776,6 +816,7 @@ class DeviceState:
                         prev_m,
                     )
                 del self._deferred_restores[gpu_index]
-            except RepartitionRefused:
-                pass  # raced a new prepare; keep deferred
+            except Exception:
+                pass  # raced a new prepare or transient HAL failure:
+                      # keep the entry, retried on the next unprepare
         return restored

@@ -242,7 +242,30 @@ class DemoHarness:
                     return False
             return True
 
-        _wait(all_allocated, 30, "controller allocation")
+        try:
+            _wait(all_allocated, 30, "controller allocation")
+        except TimeoutError:
+            pending = [
+                name
+                for name in claim_uid_by_key
+                if not (
+                    self.api.store.get_resource_claim("demo", name).get(
+                        "status"
+                    )
+                    or {}
+                ).get("allocation")
+            ]
+            n_dev = sum(
+                len(s["spec"]["devices"])
+                for s in self.api.store.list_resource_slices(DRIVER_NAME)
+            )
+            print(
+                f"[demo] FAILED: claim(s) {pending} not allocatable on "
+                f"this node ({n_dev} published device(s)) — the spec "
+                "likely needs more/other GPUs than this node has "
+                "(e.g. gpu-test1 needs 2 whole GPUs)"
+            )
+            return 1
         print(f"[demo] {len(claim_uid_by_key)} claim(s) allocated by the controller")
 
         # kubelet role: prepare each pod's claims, print the pod view

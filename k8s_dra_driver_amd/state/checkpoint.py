@@ -68,8 +68,17 @@ class PreparedClaim:
 
     @classmethod
     def from_v1(cls, d: dict) -> "PreparedClaim":
-        devs = [PreparedDevice(**pd) for pd in d.pop("devices", [])]
-        return cls(devices=devs, **{k: v for k, v in d.items()})
+        import dataclasses
+
+        dev_fields = {f.name for f in dataclasses.fields(PreparedDevice)}
+        devs = [
+            PreparedDevice(**{k: v for k, v in pd.items() if k in dev_fields})
+            for pd in d.pop("devices", [])
+        ]
+        known = {f.name for f in dataclasses.fields(cls)}
+        # unknown fields (written by a newer same-schema plugin) are
+        # dropped rather than fatal — forward-compatible reads
+        return cls(devices=devs, **{k: v for k, v in d.items() if k in known})
 
 
 def _checksum(payload: dict) -> int:
@@ -102,10 +111,18 @@ class CheckpointStore:
         safe = claim_uid.replace("/", "_")
         return os.path.join(self.claims_dir, f"{safe}.json")
 
+    #: on-disk schema version; bump only with a migration in _read_disk
+    SCHEMA_VERSION = 1
+
     def write(self, claim: PreparedClaim) -> None:
         v1 = claim.to_v1()
         atomic_write_json(
-            self._path(claim.claim_uid), {"checksum": _checksum(v1), "v1": v1}
+            self._path(claim.claim_uid),
+            {
+                "version": self.SCHEMA_VERSION,
+                "checksum": _checksum(v1),
+                "v1": v1,
+            },
         )
         with self._lock:
             self._cache[claim.claim_uid] = claim
@@ -137,6 +154,13 @@ class CheckpointStore:
             return None
         except (json.JSONDecodeError, OSError) as e:
             raise CheckpointCorrupt(f"{path}: unreadable: {e}") from e
+        version = obj.get("version", 1)  # round-1 files carry no field
+        if version > self.SCHEMA_VERSION:
+            raise CheckpointCorrupt(
+                f"{path}: checkpoint schema v{version} is newer than this "
+                f"plugin understands (v{self.SCHEMA_VERSION}) — refusing to "
+                "guess (downgrade rollback safety)"
+            )
         v1 = obj.get("v1")
         if v1 is None or obj.get("checksum") != _checksum(v1):
             raise CheckpointCorrupt(f"{path}: checksum mismatch")

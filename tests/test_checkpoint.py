@@ -94,3 +94,43 @@ def test_recovery_scan(tmp_path):
     recovered = fresh.list_all()
     assert sorted(recovered) == ["a", "b", "c"]
     assert recovered["b"].devices[0].pool_name == "node-1"
+
+
+def test_schema_versioning(tmp_path):
+    """Upgrade safety: round-1 files (no version field) read fine; files
+    from a NEWER schema are refused loudly; unknown fields from a newer
+    same-schema plugin are tolerated."""
+    store = CheckpointStore(str(tmp_path))
+    store.write(_claim())
+    path = os.path.join(str(tmp_path), "claims", "uid-1.json")
+    obj = json.load(open(path))
+    assert obj["version"] == CheckpointStore.SCHEMA_VERSION
+
+    # round-1 file: no version field -> treated as v1
+    del obj["version"]
+    json.dump(obj, open(path, "w"))
+    assert CheckpointStore(str(tmp_path)).read("uid-1") is not None
+
+    # future schema -> loud refusal
+    obj["version"] = 99
+    json.dump(obj, open(path, "w"))
+    with pytest.raises(CheckpointCorrupt, match="newer than this plugin"):
+        CheckpointStore(str(tmp_path)).read("uid-1")
+
+
+def test_unknown_fields_tolerated(tmp_path):
+    store = CheckpointStore(str(tmp_path))
+    store.write(_claim())
+    path = os.path.join(str(tmp_path), "claims", "uid-1.json")
+    obj = json.load(open(path))
+    # a newer plugin added fields within the same schema version
+    obj["v1"]["future_field"] = {"x": 1}
+    obj["v1"]["devices"][0]["future_dev_field"] = True
+    obj["checksum"] = None  # recompute below
+    import zlib as _zlib
+
+    data = json.dumps(obj["v1"], sort_keys=True, separators=(",", ":")).encode()
+    obj["checksum"] = _zlib.crc32(data) & 0xFFFFFFFF
+    json.dump(obj, open(path, "w"))
+    pc = CheckpointStore(str(tmp_path)).read("uid-1")
+    assert pc is not None and pc.devices[0].device_name == "gpu-0"

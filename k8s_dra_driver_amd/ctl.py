@@ -69,6 +69,7 @@ def cmd_topology(args) -> int:
 def cmd_slice(args) -> int:
     lib = _lib(args)
     devices = []
+    shared_counters = []
     for g in lib.enumerate():
         if g.partitions:
             devices.extend(
@@ -76,8 +77,35 @@ def cmd_slice(args) -> int:
                 for p in g.partitions
             )
         else:
-            devices.append(AllocatableDevice.from_gpu(g).to_device())
-    print(json.dumps(devices, indent=2, sort_keys=True))
+            if getattr(args, "prospective", ""):
+                from .hal.model import (
+                    gpu_device_with_counters,
+                    prospective_partition_devices,
+                    shared_counter_set,
+                )
+                from .partition.catalog import (
+                    make_profile,
+                    preferred_memory_mode,
+                )
+
+                shared_counters.append(shared_counter_set(g))
+                devices.append(gpu_device_with_counters(g))
+                prof = make_profile(
+                    args.prospective.upper(),
+                    preferred_memory_mode(
+                        args.prospective.upper(), g.nps_caps
+                    ),
+                    vram_total_mib=g.vram_total_mib or 288 * 1024,
+                    cu_count=g.cu_count or 256,
+                    nps_caps=g.nps_caps,
+                )
+                devices.extend(prospective_partition_devices(g, prof))
+            else:
+                devices.append(AllocatableDevice.from_gpu(g).to_device())
+    out: dict = {"devices": devices}
+    if shared_counters:
+        out["sharedCounters"] = shared_counters
+    print(json.dumps(out, indent=2, sort_keys=True))
     return 0
 
 
@@ -143,7 +171,14 @@ def main(argv=None) -> int:
     sub = ap.add_subparsers(dest="cmd", required=True)
     sub.add_parser("list")
     sub.add_parser("topology")
-    sub.add_parser("slice")
+    sp = sub.add_parser("slice")
+    sp.add_argument(
+        "--prospective",
+        default="",
+        choices=["", "cpx", "dpx", "qpx"],
+        help="include prospective partitions + counter sets (what "
+        "--prospective-partitions would publish)",
+    )
     pp = sub.add_parser("partition")
     pp.add_argument("gpu", type=int)
     pp.add_argument("compute")

@@ -21,8 +21,8 @@ def test_topology_matrix(capsys):
 
 def test_slice_json(capsys):
     assert main(["--hal", "fake", "slice"]) == 0
-    devices = json.loads(capsys.readouterr().out)
-    assert len(devices) == 8
+    out = json.loads(capsys.readouterr().out)
+    assert len(out["devices"]) == 8
     assert devices[0]["basic"]["attributes"]["gpu.amd.com/type"]["string"] == "gpu"
 
 
@@ -61,3 +61,10 @@ def test_ctl_profile_fetches_diag(tmp_path):
         assert buf.getvalue().startswith("# cpu profile")
     finally:
         srv.stop()
+
+
+def test_slice_prospective(capsys):
+    assert main(["--hal", "fake", "slice", "--prospective", "cpx"]) == 0
+    out = json.loads(capsys.readouterr().out)
+    assert len(out["devices"]) == 8 + 64
+    assert len(out["sharedCounters"]) == 8

@@ -23,7 +23,10 @@ def test_slice_json(capsys):
     assert main(["--hal", "fake", "slice"]) == 0
     out = json.loads(capsys.readouterr().out)
     assert len(out["devices"]) == 8
-    assert devices[0]["basic"]["attributes"]["gpu.amd.com/type"]["string"] == "gpu"
+    assert (
+        out["devices"][0]["basic"]["attributes"]["gpu.amd.com/type"]["string"]
+        == "gpu"
+    )
 
 
 def test_partition_and_list(capsys):

@@ -17,6 +17,7 @@ from __future__ import annotations
 
 from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
 
+DRA_V1_PKG = "k8s.io.kubelet.pkg.apis.dra.v1"
 DRA_V1BETA1_PKG = "k8s.io.kubelet.pkg.apis.dra.v1beta1"
 DRA_V1ALPHA_PKG = "v1alpha3"
 REG_PKG = "pluginregistration"
@@ -164,6 +165,7 @@ def _add_registration_file() -> None:
     _pool.Add(f)
 
 
+_add_dra_file(DRA_V1_PKG, "dra_v1.proto")
 _add_dra_file(DRA_V1BETA1_PKG, "dra_v1beta1.proto")
 _add_dra_file(DRA_V1ALPHA_PKG, "dra_v1alpha4.proto")
 _add_registration_file()
@@ -197,10 +199,12 @@ class DraMessages:
         return f"{self.package}.{self.service}"
 
 
+V1 = DraMessages(DRA_V1_PKG, "DRAPlugin")
 V1BETA1 = DraMessages(DRA_V1BETA1_PKG, "DRAPlugin")
 V1ALPHA4 = DraMessages(DRA_V1ALPHA_PKG, "Node")
 
 #: kubelet DRA API version strings advertised at registration
+DRA_VERSION_V1 = "v1"
 DRA_VERSION_V1BETA1 = "v1beta1"
 DRA_VERSION_V1ALPHA4 = "v1alpha4"
 

@@ -22,9 +22,11 @@ import grpc
 from .. import DRIVER_NAME
 from .driver import ClaimRef, Driver
 from .proto import (
+    DRA_VERSION_V1,
     DRA_VERSION_V1ALPHA4,
     DRA_VERSION_V1BETA1,
     REGISTRATION,
+    V1,
     V1ALPHA4,
     V1BETA1,
     DraMessages,
@@ -156,11 +158,16 @@ class PluginServer:
         )
         server.add_generic_rpc_handlers(
             (
+                _dra_handlers(self.driver, V1),
                 _dra_handlers(self.driver, V1BETA1),
                 _dra_handlers(self.driver, V1ALPHA4),
                 _registration_handlers(
                     self.plugin_sock,
-                    [DRA_VERSION_V1BETA1, DRA_VERSION_V1ALPHA4],
+                    [
+                        DRA_VERSION_V1,
+                        DRA_VERSION_V1BETA1,
+                        DRA_VERSION_V1ALPHA4,
+                    ],
                 ),
             )
         )

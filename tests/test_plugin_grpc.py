@@ -88,7 +88,7 @@ def test_registration_get_info(stack, tmp_path):
     assert info.type == "DRAPlugin"
     assert info.name == DRIVER_NAME
     assert info.endpoint == server.plugin_sock
-    assert list(info.supported_versions) == ["v1beta1", "v1alpha4"]
+    assert list(info.supported_versions) == ["v1", "v1beta1", "v1alpha4"]
     reg_channel.close()
 
 
@@ -298,3 +298,46 @@ def test_socket_watchdog_rebinds(stack):
     )(REGISTRATION.InfoRequest(), timeout=5)
     assert info.name == DRIVER_NAME
     reg_channel.close()
+
+
+def test_v1_service_prepares(stack):
+    """K8s 1.34 kubelets speak the GA dra.v1 service; same wire shapes."""
+    import grpc as _grpc
+
+    from k8s_dra_driver_amd.plugin.proto import V1
+
+    lib, kube, driver, server, _channel = stack
+    kube.put_resource_claim(
+        {
+            "metadata": {"namespace": "d", "name": "cv1", "uid": "uid-v1"},
+            "status": {
+                "allocation": {
+                    "devices": {
+                        "results": [
+                            {
+                                "request": "gpu",
+                                "driver": DRIVER_NAME,
+                                "pool": NODE,
+                                "device": "gpu-1",
+                            }
+                        ]
+                    }
+                }
+            },
+        }
+    )
+    channel = _grpc.insecure_channel(f"unix://{server.plugin_sock}")
+    m = V1
+    assert m.service_name == "k8s.io.kubelet.pkg.apis.dra.v1.DRAPlugin"
+    prepare = channel.unary_unary(
+        f"/{m.service_name}/NodePrepareResources",
+        request_serializer=lambda x: x.SerializeToString(),
+        response_deserializer=m.NodePrepareResourcesResponse.FromString,
+    )
+    req = m.NodePrepareResourcesRequest()
+    c = req.claims.add()
+    c.namespace, c.name, c.uid = "d", "cv1", "uid-v1"
+    resp = prepare(req, timeout=10)
+    assert resp.claims["uid-v1"].error == ""
+    assert resp.claims["uid-v1"].devices[0].device_name == "gpu-1"
+    channel.close()

@@ -96,6 +96,8 @@ class ResourceSlicePublisher:
         self._watch = None
         #: observability: count of heals performed (tests/metrics)
         self.heal_count = 0
+        #: optional callback per heal (metrics wiring)
+        self.on_heal = None
         #: negotiated resource.k8s.io version (resolved on first publish)
         self._api_version: Optional[str] = None
 
@@ -354,6 +356,11 @@ class ResourceSlicePublisher:
                     etype,
                 )
                 self.heal_count += 1
+                if self.on_heal is not None:
+                    try:
+                        self.on_heal()
+                    except Exception:
+                        log.debug("heal callback failed", exc_info=True)
                 devs, cdevs, scnt = self._last_publish
                 self._publish_locked(
                     list(devs),

@@ -82,6 +82,11 @@ class PluginMetrics:
             "Shared-GPU processes caught with stripped/altered CU masks",
             registry=self.registry,
         )
+        self.slice_heals = Counter(
+            "dra_resourceslice_heals_total",
+            "ResourceSlice republications triggered by external drift",
+            registry=self.registry,
+        )
 
     @contextlib.contextmanager
     def time_prepare(self):

@@ -128,6 +128,7 @@ class Driver:
             node_name=node_name,
             node_uid=node_uid,
         )
+        self.publisher.on_heal = self.metrics.slice_heals.inc
         self._pool = ThreadPoolExecutor(
             max_workers=max_concurrent_claims, thread_name_prefix="claim"
         )

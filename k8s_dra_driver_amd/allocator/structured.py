@@ -101,6 +101,37 @@ def _counter_keys(device: dict) -> frozenset:
     return frozenset(out)
 
 
+def _tolerated(request: dict, device: dict) -> bool:
+    """Device-taint check (DRA device taints, K8s 1.33): a device with a
+    NoSchedule taint is allocatable only to requests carrying a matching
+    toleration (operator Exists/Equal, optional effect scoping)."""
+    taints = device.get("taints") or []
+    if not taints:
+        return True
+    tolerations = request.get("tolerations") or []
+    for taint in taints:
+        if taint.get("effect") not in (None, "", "NoSchedule", "NoExecute"):
+            continue
+        ok = False
+        for tol in tolerations:
+            if tol.get("effect") and tol["effect"] != taint.get("effect"):
+                continue
+            op = tol.get("operator", "Equal")
+            if op == "Exists":
+                if not tol.get("key") or tol["key"] == taint.get("key"):
+                    ok = True
+                    break
+            else:  # Equal
+                if tol.get("key") == taint.get("key") and tol.get(
+                    "value", ""
+                ) == taint.get("value", ""):
+                    ok = True
+                    break
+        if not ok:
+            return False
+    return True
+
+
 def _counters_of_names(devices, names) -> set:
     out = set()
     for d in devices:
@@ -208,7 +239,13 @@ class Allocator:
             # adminAccess requests see every device, in-use included
             # (monitoring claims don't consume exclusivity)
             pool_devices = devices if r.get("adminAccess") else avail
-            cands = self.candidates_for_request(r, pool_devices)
+            cands = [
+                d
+                for d in self.candidates_for_request(r, pool_devices)
+                # adminAccess (monitoring) requests bypass taints too —
+                # reaching sick devices is their purpose
+                if r.get("adminAccess") or _tolerated(r, d)
+            ]
             mode = r.get("allocationMode", "ExactCount")
             count = len(cands) if mode == "All" else int(r.get("count", 1))
             if mode != "All" and len(cands) < count:

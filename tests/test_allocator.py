@@ -258,3 +258,94 @@ def test_combination_cap_is_logged(caplog):
     with caplog.at_level(logging.WARNING, logger="k8s_dra_driver_amd.allocator.structured"):
         Allocator().allocate(spec, devices, pool="p")
     assert any("combination cap" in r.message for r in caplog.records)
+
+
+class TestDeviceTaints:
+    """DRA device taints: NoSchedule devices need a matching toleration;
+    adminAccess (monitoring) bypasses."""
+
+    def _devs(self):
+        def d(name, taints=None):
+            out = {
+                "name": name,
+                "basic": {
+                    "attributes": {"gpu.amd.com/type": {"string": "gpu"}},
+                    "capacity": {},
+                },
+            }
+            if taints:
+                out["taints"] = taints
+            return out
+
+        sick = [{"key": "gpu.amd.com/unhealthy", "effect": "NoSchedule"}]
+        return [d("gpu-0", sick), d("gpu-1")]
+
+    def test_tainted_device_skipped(self):
+        from k8s_dra_driver_amd.allocator.structured import Allocator
+
+        spec = {
+            "devices": {
+                "requests": [{"name": "g", "deviceClassName": "gpu.amd.com"}]
+            }
+        }
+        res = Allocator().allocate(spec, self._devs(), pool="p")
+        assert res[0].device == "gpu-1"
+
+    def test_toleration_admits(self):
+        from k8s_dra_driver_amd.allocator.structured import Allocator
+
+        spec = {
+            "devices": {
+                "requests": [
+                    {
+                        "name": "g",
+                        "deviceClassName": "gpu.amd.com",
+                        "count": 2,
+                        "tolerations": [
+                            {
+                                "key": "gpu.amd.com/unhealthy",
+                                "operator": "Exists",
+                            }
+                        ],
+                    }
+                ]
+            }
+        }
+        res = Allocator().allocate(spec, self._devs(), pool="p")
+        assert sorted(r.device for r in res) == ["gpu-0", "gpu-1"]
+
+    def test_admin_access_bypasses_taints(self):
+        from k8s_dra_driver_amd.allocator.structured import Allocator
+
+        spec = {
+            "devices": {
+                "requests": [
+                    {
+                        "name": "mon",
+                        "deviceClassName": "gpu.amd.com",
+                        "count": 2,
+                        "adminAccess": True,
+                    }
+                ]
+            }
+        }
+        res = Allocator().allocate(spec, self._devs(), pool="p")
+        assert len(res) == 2
+
+    def test_untolerated_two_needed_fails(self):
+        import pytest as _pytest
+
+        from k8s_dra_driver_amd.allocator.structured import (
+            AllocationError,
+            Allocator,
+        )
+
+        spec = {
+            "devices": {
+                "requests": [
+                    {"name": "g", "deviceClassName": "gpu.amd.com", "count": 2}
+                ]
+            }
+        }
+        with _pytest.raises(AllocationError):
+            Allocator().allocate(spec, self._devs(), pool="p")

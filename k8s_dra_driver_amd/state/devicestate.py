@@ -661,6 +661,19 @@ class DeviceState:
                         gpu_index_s,
                         e,
                     )
+                except Exception as e:
+                    # revert itself failed (transient HAL error): remember
+                    # the target mode and retry when the GPU drains
+                    self._deferred_restores.setdefault(
+                        int(gpu_index_s), (modes[0], modes[1])
+                    )
+                    reverted = True  # hardware may have moved: re-sync
+                    log.warning(
+                        "rollback: revert of gpu-%s failed (%s); restore "
+                        "deferred",
+                        gpu_index_s,
+                        e,
+                    )
             if reverted:
                 self.refresh_allocatable()
                 self.write_base_cdi_spec()

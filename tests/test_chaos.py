@@ -158,8 +158,16 @@ def test_chaos_storm_leaves_node_clean(tmp_path, seed):
             for i in range(60):
                 uid = f"c{tid}-{i}"
                 gpu = rng.randrange(8)
-                flavor = rng.choice(["plain", "ts", "shared", "carve"])
-                claim = _claim(uid, f"gpu-{gpu}", flavor)
+                flavor = rng.choice(
+                    ["plain", "ts", "shared", "carve", "autocarve"]
+                )
+                if flavor == "autocarve":
+                    # scheduler-driven: claim a (possibly prospective)
+                    # partition directly; prepare carves on demand
+                    dev = f"gpu-{gpu}-cpx-{rng.randrange(8)}"
+                    claim = _claim(uid, dev, "plain")
+                else:
+                    claim = _claim(uid, f"gpu-{gpu}", flavor)
                 try:
                     state.prepare(claim)
                 except PrepareError:

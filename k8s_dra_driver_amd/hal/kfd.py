@@ -90,6 +90,8 @@ class KfdDeviceLib(DeviceLib):
                 index=index,
                 uuid=uuid,
                 oam_id=index,
+                timeslice_effective=False,
+                repartition_capable=False,
                 product_name="AMD Instinct (KFD)",
                 architecture=node.gfx_arch or "unknown",
                 pcie_bdf=node.bdf,
@@ -127,6 +129,12 @@ class KfdDeviceLib(DeviceLib):
         return out
 
     # -- partitioning: read-only in this backend ----------------------------
+    def dynamic_repartition_capable(self) -> bool:
+        return False  # read-only backend: never advertise carves
+
+    def timeslice_effective(self) -> bool:
+        return False  # requests are recorded only in KFD-only mode
+
     def set_compute_partition(self, gpu_index: int, mode: str) -> None:
         raise HalNotSupported(
             "partition control requires the amdsmi backend (KFD-only mode "

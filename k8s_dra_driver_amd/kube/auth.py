@@ -334,14 +334,7 @@ def load_in_cluster(env: Optional[dict] = None) -> KubeConnection:
     return conn
 
 
-def load_kubeconfig(
-    path: str, context: Optional[str] = None
-) -> KubeConnection:
-    """Parse a kubeconfig into a connection (clientcmd analog).
-
-    Supports inline *-data fields, file paths (resolved relative to the
-    kubeconfig), token/tokenFile, basic auth, and exec credential plugins.
-    """
+def _load_one(path: str) -> dict:
     try:
         with open(path) as f:
             cfg = yaml.safe_load(f) or {}
@@ -353,22 +346,89 @@ def load_kubeconfig(
         raise KubeAuthError(
             f"kubeconfig root must be a mapping, got {type(cfg).__name__}"
         )
+    return cfg
 
+
+def _merge_kubeconfigs(paths: List[str]) -> tuple:
+    """clientcmd KUBECONFIG-list merge: for each named entry the FIRST
+    file wins; current-context comes from the first file that sets one.
+    Returns (merged_cfg, base_dir_by_entry_name) so relative file paths
+    still resolve against the file that defined them."""
+    merged: dict = {"contexts": [], "clusters": [], "users": []}
+    bases: Dict[tuple, str] = {}
+    for p in paths:
+        cfg = _load_one(p)
+        base = os.path.dirname(os.path.abspath(p))
+        if not merged.get("current-context") and cfg.get("current-context"):
+            merged["current-context"] = cfg["current-context"]
+        for kind in ("contexts", "clusters", "users"):
+            have = {
+                e.get("name")
+                for e in merged[kind]
+                if isinstance(e, dict)
+            }
+            for e in cfg.get(kind) or []:
+                if isinstance(e, dict) and e.get("name") not in have:
+                    merged[kind].append(e)
+                    bases[(kind, e.get("name"))] = base
+    return merged, bases
+
+
+def load_kubeconfig(
+    path: str, context: Optional[str] = None
+) -> KubeConnection:
+    """Parse a kubeconfig into a connection (clientcmd analog).
+
+    Supports inline *-data fields, file paths (resolved relative to the
+    kubeconfig), token/tokenFile, basic auth, exec credential plugins,
+    and a colon-separated KUBECONFIG path list (first-wins merge).
+    """
+    if ":" in path and not os.path.exists(path):
+        paths = [p for p in path.split(":") if p]
+        cfg, bases = _merge_kubeconfigs(paths)
+        return _connection_from_cfg(cfg, context, bases)
+    cfg = _load_one(path)
+
+    base = os.path.dirname(os.path.abspath(path))
+    return _connection_from_cfg(cfg, context, None, default_base=base)
+
+
+def _connection_from_cfg(
+    cfg: dict,
+    context: Optional[str],
+    bases: Optional[Dict[tuple, str]],
+    default_base: str = "",
+) -> KubeConnection:
     ctx_name = context or cfg.get("current-context")
     if not ctx_name or not isinstance(ctx_name, str):
-        raise KubeAuthError(f"kubeconfig {path} has no current-context")
+        raise KubeAuthError("kubeconfig has no current-context")
     ctx = _named(cfg.get("contexts"), ctx_name, "context")
     cluster = _named(cfg.get("clusters"), ctx.get("cluster"), "cluster")
     user = _named(cfg.get("users"), ctx.get("user"), "user") if ctx.get("user") else {}
 
-    base = os.path.dirname(os.path.abspath(path))
+    cluster_base = (
+        bases.get(("clusters", ctx.get("cluster")), default_base)
+        if bases
+        else default_base
+    )
+    user_base = (
+        bases.get(("users", ctx.get("user")), default_base)
+        if bases
+        else default_base
+    )
 
-    def respath(p) -> Optional[str]:
+    def _respath(p, base) -> Optional[str]:
         if not p:
             return None
         if not isinstance(p, str):
             raise KubeAuthError("kubeconfig file path fields must be strings")
         return p if os.path.isabs(p) else os.path.join(base, p)
+
+    def respath(p) -> Optional[str]:
+        return _respath(p, user_base)
+
+    def respath_cluster(p) -> Optional[str]:
+        return _respath(p, cluster_base)
 
     ca_data = None
     if cluster.get("certificate-authority-data"):
@@ -376,7 +436,7 @@ def load_kubeconfig(
     conn = KubeConnection(
         cluster.get("server", ""),
         insecure_skip_verify=bool(cluster.get("insecure-skip-tls-verify")),
-        ca_file=respath(cluster.get("certificate-authority")),
+        ca_file=respath_cluster(cluster.get("certificate-authority")),
         ca_data=ca_data,
     )
 

@@ -525,3 +525,52 @@ def test_in_cluster_happy_path(tmp_path, pki):
 def test_not_in_cluster_no_kubeconfig_is_loud():
     with pytest.raises(KubeAuthError, match="KUBERNETES_SERVICE_HOST"):
         load_in_cluster({})
+
+
+def test_kubeconfig_list_merge(tmp_path):
+    """Colon-separated KUBECONFIG list: first-wins merge, relative cert
+    paths resolve against the file that defined the entry (clientcmd)."""
+    d1 = tmp_path / "a"
+    d2 = tmp_path / "b"
+    d1.mkdir()
+    d2.mkdir()
+    (d1 / "ca.crt").write_text("AAA")
+    kc1 = d1 / "kc1"
+    kc1.write_text(
+        json.dumps(
+            {
+                "clusters": [
+                    {
+                        "name": "c1",
+                        "cluster": {
+                            "server": "https://one",
+                            "certificate-authority": "ca.crt",
+                        },
+                    }
+                ],
+                "contexts": [],
+                "users": [{"name": "u1", "user": {"token": "t1"}}],
+            }
+        )
+    )
+    kc2 = d2 / "kc2"
+    kc2.write_text(
+        json.dumps(
+            {
+                "current-context": "ctx2",
+                "clusters": [
+                    {"name": "c1", "cluster": {"server": "https://SHADOWED"}},
+                    {"name": "c2", "cluster": {"server": "https://two"}},
+                ],
+                "contexts": [
+                    {"name": "ctx2", "context": {"cluster": "c1", "user": "u1"}}
+                ],
+                "users": [{"name": "u2", "user": {"token": "t2"}}],
+            }
+        )
+    )
+    conn = load_kubeconfig(f"{kc1}:{kc2}")
+    # cluster c1 from the FIRST file wins (server + relative CA path)
+    assert conn.server == "https://one"
+    assert conn.ca_file == str(d1 / "ca.crt")
+    assert conn.headers() == {"Authorization": "Bearer t1"}

@@ -19,6 +19,13 @@ def _env(name: str, default: str = "") -> str:
 
 def build_parser() -> argparse.ArgumentParser:
     p = argparse.ArgumentParser("amd-dra-controller")
+    from ..version import version_string
+
+    p.add_argument(
+        "--version",
+        action="version",
+        version=f"%(prog)s {version_string()}",
+    )
     p.add_argument("--kubeconfig", default=_env("KUBECONFIG", ""))
     p.add_argument(
         "--poll-interval", type=float, default=float(_env("POLL_INTERVAL", "10"))

@@ -33,6 +33,13 @@ def _env(name: str, default: str = "") -> str:
 
 def build_parser() -> argparse.ArgumentParser:
     p = argparse.ArgumentParser("amd-dra-kubeletplugin")
+    from ..version import version_string
+
+    p.add_argument(
+        "--version",
+        action="version",
+        version=f"%(prog)s {version_string()}",
+    )
     p.add_argument(
         "--node-name",
         default=_env("NODE_NAME", os.uname().nodename),

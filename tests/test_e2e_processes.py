@@ -314,3 +314,56 @@ def test_plugin_process_mtls_apiserver(tmp_path):
             proc.kill()
             proc.wait(timeout=5)
         srv.stop()
+
+
+@pytest.mark.timeout(90)
+def test_plugin_process_publishes_v1beta2(tmp_path):
+    """Real plugin binary against an apiserver preferring v1beta2: the
+    published slices carry the flattened Device shape + counters when
+    prospective partitioning is on."""
+    from k8s_dra_driver_amd.kube.miniapiserver import MiniApiServer
+
+    srv = MiniApiServer()
+    srv.store.api_versions = ["v1beta2", "v1beta1"]
+    srv.start()
+    srv.store.put_node({"metadata": {"name": "e2e-node"}})
+    kubeconfig = srv.write_kubeconfig(str(tmp_path / "kubeconfig"))
+    env = dict(
+        os.environ, PYTHONPATH=REPO, NODE_NAME="e2e-node", KUBECONFIG=kubeconfig
+    )
+    proc = subprocess.Popen(
+        [
+            sys.executable,
+            "-m",
+            "k8s_dra_driver_amd.plugin.main",
+            "--hal", "fake",
+            "--prospective-partitions", "cpx",
+            "--cdi-root", str(tmp_path / "cdi"),
+            "--plugin-path", str(tmp_path / "plugins" / DRIVER_NAME),
+            "--plugin-registration-path", str(tmp_path / "plugins_registry"),
+        ],
+        env=env,
+        cwd=REPO,
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        text=True,
+    )
+    try:
+        _wait_for(
+            lambda: srv.store.list_resource_slices(DRIVER_NAME),
+            what="v1beta2 slice publication",
+        )
+        slices = srv.store.list_resource_slices(DRIVER_NAME)
+        assert slices[0]["apiVersion"] == "resource.k8s.io/v1beta2"
+        devs = [d for s in slices for d in s["spec"]["devices"]]
+        assert len(devs) == 8 + 64  # whole GPUs + prospective partitions
+        assert all("basic" not in d for d in devs)
+        assert any("sharedCounters" in s["spec"] for s in slices)
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait(timeout=5)
+        srv.stop()

@@ -45,9 +45,16 @@ def labels_for_node(devices: List[dict]) -> Dict[str, str]:
     """Derive node labels from that node's published devices."""
     gpus = [d for d in devices if _attr(d, "type") == "gpu"]
     parts = [d for d in devices if _attr(d, "type") == "partition"]
+    # physical GPU count = distinct dies, whether published as a whole
+    # GPU, as partitions, or BOTH (prospective-partition mode publishes
+    # the whole GPU alongside its would-be partitions)
+    dies = {_attr(d, "uuid") for d in gpus} | {
+        _attr(p, "parentUUID") for p in parts
+    }
+    dies.discard(None)
     labels: Dict[str, str] = {
         f"{LABEL_PREFIX}/gpu.present": "true" if devices else "false",
-        f"{LABEL_PREFIX}/gpu.count": str(len(gpus) + len({_attr(p, "parentUUID") for p in parts})),
+        f"{LABEL_PREFIX}/gpu.count": str(len(dies)),
         f"{LABEL_PREFIX}/device.count": str(len(devices)),
     }
     archs = {_attr(d, "architecture") for d in devices if _attr(d, "architecture")}

@@ -91,3 +91,38 @@ def test_labels_removed_when_slices_gone():
     mgr.reconcile_once()
     labels = kube.get_node("node-a")["metadata"]["labels"]
     assert "gpu.amd.com/gpu.count" not in labels
+
+
+def test_gpu_count_with_prospective_publication(tmp_path):
+    """gpu.count must count DIES, not devices: prospective-partition mode
+    publishes each whole GPU alongside its would-be partitions."""
+    from k8s_dra_driver_amd import DRIVER_NAME
+    from k8s_dra_driver_amd.controller.manager import labels_for_node
+    from k8s_dra_driver_amd.hal import FakeDeviceLib
+    from k8s_dra_driver_amd.kube.client import InMemoryKube
+    from k8s_dra_driver_amd.plugin.driver import Driver
+
+    lib = FakeDeviceLib()
+    lib.open()
+    kube = InMemoryKube()
+    kube.api_versions = ["v1beta2", "v1beta1"]
+    d = Driver(
+        lib,
+        kube,
+        node_name="n",
+        cdi_root=str(tmp_path / "cdi"),
+        checkpoint_root=str(tmp_path / "state"),
+        use_tmpfs=False,
+        prospective_partitions="cpx",
+    )
+    d.startup()
+    devices = [
+        dev
+        for s in kube.list_resource_slices(DRIVER_NAME)
+        for dev in s["spec"]["devices"]
+    ]
+    assert len(devices) == 72  # 8 whole + 64 prospective
+    labels = labels_for_node(devices)
+    assert labels["gpu.amd.com/gpu.count"] == "8"
+    assert labels["gpu.amd.com/device.count"] == "72"
+    d.shutdown(unpublish=False)

@@ -135,3 +135,21 @@ def test_kubeconfig_loader_never_crashes(tmp_path_factory, doc):
         conn.ssl_verify()
     except KubeAuthError:
         pass  # the typed error is the contract
+
+
+@settings(max_examples=100, deadline=None)
+@given(doc1=_yamlish, doc2=_yamlish)
+def test_kubeconfig_merge_never_crashes(tmp_path_factory, doc1, doc2):
+    import yaml as _yaml
+
+    from k8s_dra_driver_amd.kube.auth import KubeAuthError, load_kubeconfig
+
+    d = tmp_path_factory.mktemp("kcm")
+    p1, p2 = d / "a", d / "b"
+    p1.write_text(_yaml.safe_dump(doc1))
+    p2.write_text(_yaml.safe_dump(doc2))
+    try:
+        conn = load_kubeconfig(f"{p1}:{p2}")
+        conn.ssl_verify()
+    except KubeAuthError:
+        pass

@@ -4,9 +4,10 @@ The socket/registration lifecycle the reference inherits from the vendored
 ``kubeletplugin`` helper (``draplugin.go:280-350``): a DRA service socket at
 ``<plugin-dir>/plugin.sock`` and a registration socket at
 ``<registry-dir>/<driver-name>.sock`` where kubelet's pluginwatcher calls
-``Registration/GetInfo``. Both DRA API versions (v1beta1 DRAPlugin +
-v1alpha4 Node) are served for kubelet version negotiation, like the
-reference registers both (``draplugin.go:342-350``).
+``Registration/GetInfo``. Three DRA API versions are served for kubelet
+version negotiation — dra.v1 (GA, K8s 1.34), v1beta1 DRAPlugin and
+v1alpha4 Node — one generation past the reference's pair
+(``draplugin.go:342-350``).
 """
 
 from __future__ import annotations

@@ -140,3 +140,64 @@ def test_manager_integration():
     mgr = ControllerManager(kube, allocate_claims=True)
     mgr.reconcile_once()
     assert kube.get_resource_claim("d", "c-u1")["status"].get("allocation")
+
+
+def test_cross_node_topology_placement(tmp_path):
+    """Two simulated nodes (one plugin instance each via --gpu-indices
+    scoping, nvkind analog): a 3-GPU topology claim must land WHOLLY on
+    one node — the controller scores nodes and never splits a claim."""
+    from k8s_dra_driver_amd.hal import FakeDeviceLib, FakeNodeConfig
+    from k8s_dra_driver_amd.plugin.driver import Driver
+
+    kube = InMemoryKube()
+    drivers = []
+    # node-a manages 2 GPUs, node-b manages 4: only node-b fits count=3
+    for node, idxs in (("node-a", [0, 1]), ("node-b", [2, 3, 4, 5])):
+        lib = FakeDeviceLib(FakeNodeConfig(num_gpus=8))
+        lib.open()
+        d = Driver(
+            lib,
+            kube,
+            node_name=node,
+            cdi_root=str(tmp_path / node / "cdi"),
+            checkpoint_root=str(tmp_path / node / "state"),
+            use_tmpfs=False,
+            gpu_indices=idxs,
+        )
+        d.startup()
+        drivers.append(d)
+
+    claim = {
+        "metadata": {"namespace": "d", "name": "tri", "uid": "uid-tri"},
+        "spec": {
+            "devices": {
+                "requests": [
+                    {
+                        "name": "tri",
+                        "deviceClassName": "gpu.amd.com",
+                        "count": 3,
+                    }
+                ]
+            }
+        },
+    }
+    kube.put_resource_claim(claim)
+    sched = ClaimScheduler(kube)
+    assert sched.reconcile_once() == ["uid-tri"]
+    stored = kube.get_resource_claim("d", "tri")
+    alloc = stored["status"]["allocation"]
+    results = alloc["devices"]["results"]
+    assert len(results) == 3
+    pools = {r["pool"] for r in results}
+    assert pools == {"node-b"}  # whole claim on the only node that fits
+    # nodeSelector pins the pod to that node
+    terms = alloc["nodeSelector"]["nodeSelectorTerms"]
+    assert any(
+        any(
+            "node-b" in (m.get("values") or [])
+            for m in t.get("matchFields", []) + t.get("matchExpressions", [])
+        )
+        for t in terms
+    )
+    for d in drivers:
+        d.shutdown(unpublish=False)

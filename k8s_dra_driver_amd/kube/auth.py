@@ -367,7 +367,12 @@ def _merge_kubeconfigs(paths: List[str]) -> tuple:
                 for e in merged[kind]
                 if isinstance(e, dict)
             }
-            for e in cfg.get(kind) or []:
+            entries = cfg.get(kind) or []
+            if not isinstance(entries, list):
+                raise KubeAuthError(
+                    f"kubeconfig {p}: {kind} must be a list"
+                )
+            for e in entries:
                 if isinstance(e, dict) and e.get("name") not in have:
                     merged[kind].append(e)
                     bases[(kind, e.get("name"))] = base

@@ -193,7 +193,13 @@ class BenchRank:
             self.driver, plugin_dir=os.path.join(self.tmp, "plugin")
         )
         self.server.start()
-        self.channel = grpc.insecure_channel(f"unix://{self.server.plugin_sock}")
+        self.channel = grpc.insecure_channel(
+            f"unix://{self.server.plugin_sock}",
+            options=[
+                ("grpc.optimization_target", "latency"),
+                ("grpc.enable_retries", 0),
+            ],
+        )
         m = V1BETA1
         self.prepare = self.channel.unary_unary(
             f"/{m.service_name}/NodePrepareResources",

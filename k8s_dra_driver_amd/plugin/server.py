@@ -155,7 +155,12 @@ class PluginServer:
         server = grpc.server(
             ThreadPoolExecutor(
                 max_workers=self._grpc_workers, thread_name_prefix="grpc"
-            )
+            ),
+            options=[
+                # prepare latency is the north-star metric; kubelet is a
+                # single local client over a unix socket
+                ("grpc.optimization_target", "latency"),
+            ],
         )
         server.add_generic_rpc_handlers(
             (

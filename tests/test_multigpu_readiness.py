@@ -82,3 +82,13 @@ def test_bench_launch_shape_topo4_distributed():
     assert d["n_gpus"] == 2
     assert d["config"]["model"] == "dra-claim-lifecycle/topo4"
     assert d["value"] > 0
+
+
+@pytest.mark.timeout(280)
+def test_bench_launch_shape_autocpx_distributed():
+    """Scheduler-driven carve mode under the distributed launch: each
+    rank auto-carves its own GPU, drains, restores."""
+    d = _run_bench_launch(2, extra=("--config", "autocpx", "--hal", "fake"))
+    assert d["n_gpus"] == 2
+    assert d["config"]["model"] == "dra-claim-lifecycle/autocpx"
+    assert d["value"] > 0

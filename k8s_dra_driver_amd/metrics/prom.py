@@ -87,6 +87,11 @@ class PluginMetrics:
             "ResourceSlice republications triggered by external drift",
             registry=self.registry,
         )
+        self.deferred_restores = Gauge(
+            "dra_deferred_mode_restores",
+            "GPUs whose partition-mode restore is pending a drain",
+            registry=self.registry,
+        )
 
     @contextlib.contextmanager
     def time_prepare(self):

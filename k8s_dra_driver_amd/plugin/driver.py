@@ -129,6 +129,9 @@ class Driver:
             node_uid=node_uid,
         )
         self.publisher.on_heal = self.metrics.slice_heals.inc
+        self.state.on_deferred_restores_change = (
+            self.metrics.deferred_restores.set
+        )
         self._pool = ThreadPoolExecutor(
             max_workers=max_concurrent_claims, thread_name_prefix="claim"
         )

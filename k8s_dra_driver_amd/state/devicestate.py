@@ -122,6 +122,8 @@ class DeviceState:
         self.on_warning = None
         #: optional callback per performed mode switch (metrics)
         self.on_repartition = None
+        #: optional callback(count) when the deferred-restore set changes
+        self.on_deferred_restores_change = None
 
         self._registry_lock = threading.Lock()
         self._claim_locks: Dict[str, "_RefLock"] = {}
@@ -667,6 +669,7 @@ class DeviceState:
                     self._deferred_restores.setdefault(
                         int(gpu_index_s), (modes[0], modes[1])
                     )
+                    self._notify_deferred()
                     reverted = True  # hardware may have moved: re-sync
                     log.warning(
                         "rollback: revert of gpu-%s failed (%s); restore "
@@ -777,6 +780,7 @@ class DeviceState:
                         prev_c,
                         prev_m,
                     )
+                    self._notify_deferred()
                     log.warning(
                         "leaving gpu-%s partitioned for now (%s); restore "
                         "deferred until the GPU drains",
@@ -799,8 +803,16 @@ class DeviceState:
         actual hardware state."""
         if not e.reverted:
             self._deferred_restores.setdefault(gpu_index, e.original)
+            self._notify_deferred()
             self.refresh_allocatable()
             self.write_base_cdi_spec()
+
+    def _notify_deferred(self) -> None:
+        if self.on_deferred_restores_change is not None:
+            try:
+                self.on_deferred_restores_change(len(self._deferred_restores))
+            except Exception:
+                pass
 
     def _retry_deferred_restores(self, requesting_claim: str) -> bool:
         """Apply deferred mode restores for GPUs that have drained (the
@@ -829,6 +841,7 @@ class DeviceState:
                         prev_m,
                     )
                 del self._deferred_restores[gpu_index]
+                self._notify_deferred()
             except Exception:
                 pass  # raced a new prepare or transient HAL failure:
                       # keep the entry, retried on the next unprepare

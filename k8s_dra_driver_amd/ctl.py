@@ -150,6 +150,26 @@ def cmd_health(args) -> int:
     return rc
 
 
+def cmd_labels(args) -> int:
+    """Preview the node labels the controller would derive from this
+    node's devices (gpu.count, architecture, hive, partition modes)."""
+    from .controller.manager import labels_for_node
+
+    lib = _lib(args)
+    devices = []
+    for g in lib.enumerate():
+        if g.partitions:
+            devices.extend(
+                AllocatableDevice.from_partition(g, p).to_device()
+                for p in g.partitions
+            )
+        else:
+            devices.append(AllocatableDevice.from_gpu(g).to_device())
+    for k, v in sorted(labels_for_node(devices).items()):
+        print(f"{k}={v}")
+    return 0
+
+
 def cmd_profile(args) -> int:
     """Fetch an on-demand CPU profile from a running plugin/controller's
     diag server (/debug/profile) — the `go tool pprof` moment for the
@@ -183,6 +203,7 @@ def main(argv=None) -> int:
     pp.add_argument("gpu", type=int)
     pp.add_argument("compute")
     pp.add_argument("memory", nargs="?", default="NPS1")
+    sub.add_parser("labels")
     hp = sub.add_parser("health")
     hp.add_argument("--probe", action="store_true", help="also run HIP kernels")
     prof = sub.add_parser("profile")
@@ -196,6 +217,7 @@ def main(argv=None) -> int:
         "slice": cmd_slice,
         "partition": cmd_partition,
         "health": cmd_health,
+        "labels": cmd_labels,
         "profile": cmd_profile,
     }[args.cmd](args)
 

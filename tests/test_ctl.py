@@ -71,3 +71,10 @@ def test_slice_prospective(capsys):
     out = json.loads(capsys.readouterr().out)
     assert len(out["devices"]) == 8 + 64
     assert len(out["sharedCounters"]) == 8
+
+
+def test_labels_preview(capsys):
+    assert main(["--hal", "fake", "labels"]) == 0
+    out = capsys.readouterr().out
+    assert "gpu.amd.com/gpu.count=8" in out
+    assert "gpu.amd.com/gpu.architecture=gfx950" in out
